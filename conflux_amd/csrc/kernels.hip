@@ -1,0 +1,746 @@
+// kernels.hip — CDNA4 (gfx950) kernels of the MI355X-native CONFLUX LU engine.
+//
+// Re-expressions of the reference's compute units (SURVEY.md §2 checklist):
+//   k_dgemm_f64        <- trailing-update cblas_dgemm (conflux_opt.hpp:1628-1633)
+//                         and the panel-update GEMMs of blocked getrf/TRSM;
+//                         hand-written v_mfma_f64_16x16x4_f64, LDS-staged.
+//   k_panel_col        <- one column of LAPACKE_dgetrf partial pivoting
+//                         (conflux_opt.hpp:143-166 LUP): grid-wide first-max
+//                         argmax + row swap + scale + rank-1 on a col-major
+//                         sub-panel; agent-scope release/acquire handshake.
+//   k_trsm_*           <- cblas_dtrsm Right/Upper/NonUnit (:1347) and
+//                         Left/Lower/Unit (:1539), 32-wide diagonal blocks +
+//                         k_dgemm_f64 updates.
+//   k_row_gather/scatter, k_laswp, k_copy2d, ...
+//                      <- push_pivots_up / permute_rows / mcopy
+//                         (conflux_opt.hpp:176-218, utils.hpp:48-160,
+//                          memory_utils.hpp:8-34) as coalesced index-vector
+//                         kernels.
+//   k_init_matrix      <- lu_params::InitMatrix random fill (splitmix64
+//                         variant, oracle/gen_input.py — bit-identical).
+//
+// All fp64.  No CUDA-compat shims, no library GEMMs on the hot path.
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+#include "kernels.hpp"
+
+#define DEVFN __device__ __forceinline__
+
+typedef double f64x4 __attribute__((ext_vector_type(4)));
+
+namespace ck {
+
+// ---------------------------------------------------------------------------
+// input generator (matches oracle/gen_input.py bit-for-bit)
+// ---------------------------------------------------------------------------
+DEVFN uint64_t splitmix64(uint64_t z) {
+    z += 0x9E3779B97F4A7C15ull;
+    z ^= z >> 30;
+    z *= 0xBF58476D1CE4E5B9ull;
+    z ^= z >> 27;
+    z *= 0x94D049BB133111EBull;
+    z ^= z >> 31;
+    return z;
+}
+
+__global__ void k_init_matrix(double *__restrict__ A, int Ml, int Nl, int v,
+                              int Px, int Py, int pi, int pj, int zero_layer,
+                              uint64_t seed) {
+    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t total = (int64_t)Ml * Nl;
+    if (idx >= total) return;
+    if (zero_layer) { A[idx] = 0.0; return; }
+    const int r = (int)(idx / Nl), c = (int)(idx % Nl);
+    // local (r, c) -> global (i, j): tile-cyclic map (layout.cpp:95-123)
+    const int64_t gi = (int64_t)(r / v * Px + pi) * v + r % v;
+    const int64_t gj = (int64_t)(c / v * Py + pj) * v + c % v;
+    uint64_t key = ((uint64_t)gi << 32) ^ (uint64_t)gj;
+    key += seed * 0xBF58476D1CE4E5B9ull;
+    const uint64_t h = splitmix64(key);
+    A[idx] = 5.0 + (double)(h >> 11) * (1.0 / 9007199254740992.0);
+}
+
+// ---------------------------------------------------------------------------
+// 2D strided copies / zero / add (mcopy & reduce-combine re-expressions)
+// ---------------------------------------------------------------------------
+__global__ void k_copy2d(const double *__restrict__ src, int64_t lds,
+                         double *__restrict__ dst, int64_t ldd,
+                         int rows, int64_t cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)rows * cols) return;
+    const int64_t r = i / cols, c = i % cols;
+    dst[r * ldd + c] = src[r * lds + c];
+}
+
+__global__ void k_zero2d(double *__restrict__ dst, int64_t ldd, int rows,
+                         int64_t cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)rows * cols) return;
+    dst[(i / cols) * ldd + i % cols] = 0.0;
+}
+
+__global__ void k_add2d(const double *__restrict__ src, int64_t lds,
+                        double *__restrict__ dst, int64_t ldd, int rows,
+                        int64_t cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)rows * cols) return;
+    dst[(i / cols) * ldd + i % cols] += src[(i / cols) * lds + i % cols];
+}
+
+// dst row i <- src row idx[i]   (gather; push_pivots_up phase 1/3,
+// candidate winner permute — utils.hpp inverse_permute_rows)
+__global__ void k_row_gather(const double *__restrict__ src, int64_t lds,
+                             double *__restrict__ dst, int64_t ldd,
+                             const int *__restrict__ idx, int n_rows,
+                             int64_t cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)n_rows * cols) return;
+    const int64_t r = i / cols, c = i % cols;
+    dst[r * ldd + c] = src[(int64_t)idx[r] * lds + c];
+}
+
+// dst row idx[i] <- src row i   (scatter; push_pivots_up phase 2,
+// A01 pivot-order placement — conflux_opt.hpp:1251-1258)
+__global__ void k_row_scatter(const double *__restrict__ src, int64_t lds,
+                              double *__restrict__ dst, int64_t ldd,
+                              const int *__restrict__ idx, int n_rows,
+                              int64_t cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)n_rows * cols) return;
+    const int64_t r = i / cols, c = i % cols;
+    dst[(int64_t)idx[r] * ldd + c] = src[r * lds + c];
+}
+
+// src rows idx[i] copied to dst rows dst_idx[i] within SAME buffer is unsafe;
+// engine always stages through separate buffers (3-phase like the reference).
+
+// Apply nswap sequential row swaps (i0+s <-> i0+piv[s]) to columns
+// [c0, c1) of a row-major matrix — the dlaswp step of blocked getrf.
+// piv entries are relative to row i0 (what k_panel_col records).
+// Each thread owns one column and performs the swap sequence in order.
+__global__ void k_laswp(double *__restrict__ A, int64_t lda, int i0,
+                        const int *__restrict__ piv, int nswap, int64_t c0,
+                        int64_t c1) {
+    const int64_t c = c0 + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= c1) return;
+    for (int s = 0; s < nswap; ++s) {
+        const int64_t r1 = i0 + s, r2 = i0 + piv[s];
+        if (r1 == r2) continue;
+        const double t = A[r1 * lda + c];
+        A[r1 * lda + c] = A[r2 * lda + c];
+        A[r2 * lda + c] = t;
+    }
+}
+
+// dst row dst_idx[i] <- src row src_idx[i]; row sets must be disjoint
+// (push_pivots_up phase 2: early non-pivots into vacated late-pivot slots)
+__global__ void k_row_move(const double *__restrict__ src, int64_t lds,
+                           double *__restrict__ dst, int64_t ldd,
+                           const int *__restrict__ src_idx,
+                           const int *__restrict__ dst_idx, int n_rows,
+                           int64_t cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)n_rows * cols) return;
+    const int64_t r = i / cols, c = i % cols;
+    dst[(int64_t)dst_idx[r] * ldd + c] = src[(int64_t)src_idx[r] * lds + c];
+}
+
+// row-major (rows x cols, ld) -> col-major scratch (ldc >= rows)
+__global__ void k_cm_import(const double *__restrict__ src, int64_t lds,
+                            double *__restrict__ cm, int64_t ldc, int rows,
+                            int cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)rows * cols) return;
+    const int64_t r = i / cols, c = i % cols;   // read coalesced over c
+    cm[c * ldc + r] = src[r * lds + c];
+}
+
+__global__ void k_cm_export(const double *__restrict__ cm, int64_t ldc,
+                            double *__restrict__ dst, int64_t ldd, int rows,
+                            int cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)rows * cols) return;
+    const int64_t r = i / cols, c = i % cols;   // write coalesced over c
+    dst[r * ldd + c] = cm[c * ldc + r];
+}
+
+// ---------------------------------------------------------------------------
+// panel column step: argmax + swap + scale + rank-1 on a col-major sub-panel
+// ---------------------------------------------------------------------------
+// One launch per column c of an nb-wide sub-panel held col-major in `cm`
+// (ld = ldc, rows 0..m).  Matches LAPACK dgetrf partial pivoting: pivot =
+// first row (smallest index) with max |value| in column c among rows c..m;
+// swap full sub-panel rows c <-> piv; scale col c below diag by 1/pivot
+// (multiply by reciprocal, as dscal does); rank-1 update cols c+1..nb.
+//
+// Inter-block protocol (cdna_hip_programming.md §6 G16, R1 form):
+//   every block: phase-A local argmax over its rows -> sc1 payload stores ->
+//   vmcnt drain -> relaxed-agent flag(epoch); block 0 polls all flags,
+//   reduces with the first-max rule, performs the swap OUTSIDE the disputed
+//   region, publishes both affected rows sc1 + flag; blocks then finish from
+//   registers + the published slab (no plain re-reads of swapped data).
+#define PANEL_TPB 256
+#define PANEL_RPB 512           // rows per block (2 per thread)
+#define PANEL_NB 32             // sub-panel width == register column budget
+
+struct PanelSync {
+    // per-block candidate: |value| bits, row, flag(epoch)
+    unsigned long long cand_abs[CONFLUX_PANEL_MAX_BLOCKS];
+    int cand_row[CONFLUX_PANEL_MAX_BLOCKS];
+    unsigned int cand_flag[CONFLUX_PANEL_MAX_BLOCKS];
+    // publication: new row c content (= old pivot row) and new piv row
+    // content (= old row c), columns c..nb-1 at slot index (col - c)
+    double rowc[PANEL_NB];
+    double rowpiv[PANEL_NB];
+    int piv_row;
+    unsigned int pub_flag;
+    unsigned int err;
+};
+
+typedef unsigned int __attribute__((address_space(1))) gu32;
+typedef unsigned long long __attribute__((address_space(1))) gu64;
+#define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+
+DEVFN void st_rlx_u64(unsigned long long *p, unsigned long long v) {
+    __hip_atomic_store((gu64 *)p, v, RLX_AGENT);
+}
+DEVFN unsigned long long ld_rlx_u64(const unsigned long long *p) {
+    return __hip_atomic_load((const gu64 *)p, RLX_AGENT);
+}
+DEVFN void st_rlx_u32(unsigned int *p, unsigned int v) {
+    __hip_atomic_store((gu32 *)p, v, RLX_AGENT);
+}
+DEVFN unsigned int ld_rlx_u32(const unsigned int *p) {
+    return __hip_atomic_load((const gu32 *)p, RLX_AGENT);
+}
+DEVFN void st_rlx_f64(double *p, double v) {
+    union { double d; unsigned long long u; } x;
+    x.d = v;
+    st_rlx_u64((unsigned long long *)p, x.u);
+}
+DEVFN double ld_rlx_f64(const double *p) {
+    union { double d; unsigned long long u; } x;
+    x.u = ld_rlx_u64((const unsigned long long *)p);
+    return x.d;
+}
+DEVFN void drain_stores() { asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); }
+
+__global__ __launch_bounds__(PANEL_TPB) void k_panel_col(
+    double *__restrict__ cm, int64_t ldc, int m, int nb, int c,
+    PanelSync *__restrict__ sync, int *__restrict__ ipiv, unsigned int epoch,
+    int nblocks) {
+    const int tid = threadIdx.x;
+    const int bid = blockIdx.x;
+    // rows of column c handled by this block: [c + bid*RPB + tid*2, ...)
+    const int base = c + bid * PANEL_RPB;
+    double val[2];
+    int row[2];
+    double amax = -1.0;
+    int arow = m;  // sentinel: larger than any row
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+        const int r = base + tid + q * PANEL_TPB;
+        row[q] = r;
+        val[q] = (r < m) ? cm[(int64_t)c * ldc + r] : 0.0;
+        const double a = fabs(val[q]);
+        // first-max: strictly greater, or equal with smaller row index
+        if (r < m && (a > amax || (a == amax && r < arow))) { amax = a; arow = r; }
+    }
+    // block-level (abs, row) reduce with the same rule
+    __shared__ double s_abs[PANEL_TPB];
+    __shared__ int s_row[PANEL_TPB];
+    s_abs[tid] = amax;
+    s_row[tid] = arow;
+    __syncthreads();
+    for (int w = PANEL_TPB / 2; w > 0; w >>= 1) {
+        if (tid < w) {
+            const double oa = s_abs[tid + w];
+            const int orr = s_row[tid + w];
+            if (oa > s_abs[tid] || (oa == s_abs[tid] && orr < s_row[tid])) {
+                s_abs[tid] = oa;
+                s_row[tid] = orr;
+            }
+        }
+        __syncthreads();
+    }
+    if (tid == 0) {
+        union { double d; unsigned long long u; } a;
+        a.d = s_abs[0];
+        st_rlx_u64(&sync->cand_abs[bid], a.u);
+        st_rlx_u32((unsigned int *)&sync->cand_row[bid], (unsigned int)s_row[0]);
+        drain_stores();
+        st_rlx_u32(&sync->cand_flag[bid], epoch);
+    }
+
+    // ---- block 0 = reducer -------------------------------------------------
+    if (bid == 0) {
+        __shared__ int sh_piv;
+        if (tid == 0) {
+            double best = -1.0;
+            int bestrow = m;
+            unsigned spins = 0;
+            for (int b = 0; b < nblocks; ++b) {
+                while (ld_rlx_u32(&sync->cand_flag[b]) != epoch) {
+                    __builtin_amdgcn_s_sleep(1);
+                    if (++spins > 400000000u) { st_rlx_u32(&sync->err, 1u + c); goto bail;
+                    }
+                }
+                union { double d; unsigned long long u; } a;
+                a.u = ld_rlx_u64(&sync->cand_abs[b]);
+                const int rr = (int)ld_rlx_u32((unsigned int *)&sync->cand_row[b]);
+                if (a.d > best || (a.d == best && rr < bestrow)) { best = a.d; bestrow = rr; }
+            }
+            sh_piv = bestrow;
+            ipiv[0] = bestrow;   // sub-panel-absolute row of the pivot
+        bail:;
+        }
+        __syncthreads();
+        const int piv = sh_piv;
+        // swap rows c <-> piv.  Disputed cells (row piv, cols >= c) are NOT
+        // written here — the block owning row piv finalizes them from the
+        // published slab, so there is no write race.
+        //   row c   <- old row piv  (cols 0..nb)      [reducer, sc1]
+        //   row piv <- old row c    (cols 0..c-1)     [reducer, sc1]
+        //   slab rowc[j]   = old piv row value at col c+j  (new row c)
+        //   slab rowpiv[j] = old row c value at col c+j    (new row piv)
+        if (tid < nb) {
+            const int cc = tid;
+            const double oldc = cm[(int64_t)cc * ldc + c];
+            const double oldp = cm[(int64_t)cc * ldc + piv];
+            if (piv != c) {
+                st_rlx_f64(&cm[(int64_t)cc * ldc + c], oldp);
+                if (cc < c) st_rlx_f64(&cm[(int64_t)cc * ldc + piv], oldc);
+            }
+            if (cc >= c) {
+                st_rlx_f64(&sync->rowc[cc - c], oldp);
+                st_rlx_f64(&sync->rowpiv[cc - c], oldc);
+            }
+        }
+        if (tid == 0) st_rlx_u32((unsigned int *)&sync->piv_row, (unsigned int)piv);
+        drain_stores();
+        __syncthreads();
+        if (tid == 0) st_rlx_u32(&sync->pub_flag, epoch);
+    }
+
+    // ---- all blocks: wait for publication ---------------------------------
+    __shared__ double sh_rowc[PANEL_NB];   // pivot row (new row c), cols c..nb
+    __shared__ double sh_rowpiv[PANEL_NB]; // displaced row (new row piv)
+    __shared__ int sh_piv2;
+    if (tid == 0) {
+        unsigned spins = 0;
+        while (ld_rlx_u32(&sync->pub_flag) != epoch) {
+            __builtin_amdgcn_s_sleep(1);
+            if (++spins > 400000000u) { st_rlx_u32(&sync->err, 1000000u + c); break; }
+        }
+        sh_piv2 = (int)ld_rlx_u32((unsigned int *)&sync->piv_row);
+    }
+    __syncthreads();
+    if (tid < nb - c) {
+        sh_rowc[tid] = ld_rlx_f64(&sync->rowc[tid]);
+        sh_rowpiv[tid] = ld_rlx_f64(&sync->rowpiv[tid]);
+    }
+    __syncthreads();
+    const int piv = sh_piv2;
+    const double pivval = sh_rowc[0];
+    // LAPACK dgetrf2 scaling rule: multiply by reciprocal unless the pivot
+    // is exactly zero (singular: column left unscaled, as LAPACK does).
+    const double recip = (pivval != 0.0) ? 1.0 / pivval : 0.0;
+    const bool do_scale = pivval != 0.0;
+
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+        const int r = row[q];
+        if (r >= m || r <= c) continue;   // rows strictly below the diagonal
+        double v0 = (r == piv) ? sh_rowpiv[0] : val[q];   // post-swap value
+        const double l = do_scale ? v0 * recip : v0;
+        cm[(int64_t)c * ldc + r] = l;     // final L entry (plain store)
+        for (int cc = c + 1; cc < nb; ++cc) {
+            const double prev = (r == piv) ? sh_rowpiv[cc - c]
+                                           : cm[(int64_t)cc * ldc + r];
+            cm[(int64_t)cc * ldc + r] = prev - l * sh_rowc[cc - c];
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// TRSM diagonal-block solvers (32-wide blocks; updates via k_dgemm_f64)
+// ---------------------------------------------------------------------------
+// X (nb x N, row-major ld) <- L^{-1} X with L (nb x nb, ld ldl) unit-lower:
+// thread j owns column j; column values kept in registers.
+// cblas_dtrsm Left/Lower/NoTrans/Unit re-expression (conflux_opt.hpp:1539).
+__global__ __launch_bounds__(256) void k_trsm_left_lower_unit32(
+    const double *__restrict__ L, int64_t ldl, double *__restrict__ X,
+    int64_t ldx, int nb, int64_t N) {
+    __shared__ double sL[32][33];
+    const int tid = threadIdx.x;
+    for (int i = tid; i < nb * nb; i += 256) sL[i / nb][i % nb] = L[(i / nb) * ldl + i % nb];
+    __syncthreads();
+    const int64_t j = (int64_t)blockIdx.x * 256 + tid;
+    if (j >= N) return;
+    double x[32];
+#pragma unroll
+    for (int r = 0; r < 32; ++r) x[r] = (r < nb) ? X[r * ldx + j] : 0.0;
+#pragma unroll
+    for (int r = 1; r < 32; ++r) {
+        double acc = x[r];
+#pragma unroll
+        for (int i = 0; i < r; ++i) acc -= sL[r][i] * x[i];
+        if (r < nb) x[r] = acc;
+    }
+#pragma unroll
+    for (int r = 0; r < 32; ++r)
+        if (r < nb) X[r * ldx + j] = x[r];
+}
+
+// X (M x nb, row-major ld) <- X U^{-1} with U (nb x nb, ld ldu) upper
+// non-unit: 256 rows per block staged through LDS (coalesced), thread r owns
+// row r.  cblas_dtrsm Right/Upper/NoTrans/NonUnit (conflux_opt.hpp:1347).
+__global__ __launch_bounds__(256) void k_trsm_right_upper32(
+    const double *__restrict__ U, int64_t ldu, double *__restrict__ X,
+    int64_t ldx, int nb, int64_t M) {
+    __shared__ double sU[32][33];
+    __shared__ double sX[256][33];
+    const int tid = threadIdx.x;
+    for (int i = tid; i < nb * nb; i += 256) sU[i / nb][i % nb] = U[(i / nb) * ldu + i % nb];
+    const int64_t r0 = (int64_t)blockIdx.x * 256;
+    const int rows = (int)min((int64_t)256, M - r0);
+    // stage rows r0..r0+rows coalesced: thread t covers elements t, t+256, ...
+    for (int i = tid; i < rows * nb; i += 256) sX[i / nb][i % nb] = X[(r0 + i / nb) * ldx + i % nb];
+    __syncthreads();
+    if (tid < rows) {
+        double x[32];
+#pragma unroll
+        for (int c = 0; c < 32; ++c) x[c] = (c < nb) ? sX[tid][c] : 0.0;
+#pragma unroll
+        for (int c = 0; c < 32; ++c) {
+            double acc = x[c];
+#pragma unroll
+            for (int i = 0; i < c; ++i) acc -= x[i] * sU[i][c];
+            if (c < nb) x[c] = acc / sU[c][c];
+        }
+#pragma unroll
+        for (int c = 0; c < 32; ++c)
+            if (c < nb) sX[tid][c] = x[c];
+    }
+    __syncthreads();
+    for (int i = tid; i < rows * nb; i += 256) X[(r0 + i / nb) * ldx + i % nb] = sX[i / nb][i % nb];
+}
+
+// ---------------------------------------------------------------------------
+// MFMA fp64 GEMM:  C (M x N, ldc) -= A (M x K, lda) * B (K x N, ldb)
+// ---------------------------------------------------------------------------
+// 128x128 block tile, BK=16, 4 waves (2x2), 64x64 per wave as 4x4 fragments
+// of v_mfma_f64_16x16x4_f64.  Register-staged global->LDS with the
+// write-after-barrier placement (guide §5.5 T14); A transposed in LDS so
+// fragment reads are conflict-free; XCD-aware block swizzle (T1).
+#define GEMM_BM 128
+#define GEMM_BN 128
+#define GEMM_BK 16
+#define GEMM_TPB 256
+
+__global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
+    const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
+    int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
+    int ntm, int ntn) {
+    // bijective XCD swizzle (guide §5: q/r form)
+    int wg = blockIdx.x;
+    {
+        const int nwg = ntm * ntn;
+        const int q = nwg >> 3, r = nwg & 7;
+        const int xcd = wg & 7, idx = wg >> 3;
+        // inverse of dispatch round-robin: give each XCD a contiguous chunk
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+        if (nwg < 8) wg = blockIdx.x;   // tiny grids: identity
+    }
+    const int tm = wg / ntn, tn = wg % ntn;
+    const int row0 = tm * GEMM_BM;
+    const int64_t col0 = (int64_t)tn * GEMM_BN;
+
+    __shared__ double As[GEMM_BK][GEMM_BM + 1];   // transposed, padded
+    __shared__ double Bs[GEMM_BK][GEMM_BN + 2];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm0 = (wave >> 1) * 64;             // wave's 64x64 sub-tile
+    const int wn0 = (wave & 1) * 64;
+    const int frow = lane & 15;                   // fragment row/col lane part
+    const int fk = lane >> 4;                     // fragment k lane part
+
+    f64x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = f64x4{0, 0, 0, 0};
+
+    // staging registers: 8 A elements + 8 B elements per thread per K-tile
+    // (element e = tid + i*256 over the 2048-element tile; A reads coalesce
+    // over the 16-wide rows, B reads over the 128-wide rows)
+    double ra[8], rb[8];
+    const int ktiles = (K + GEMM_BK - 1) / GEMM_BK;
+
+    auto load_a = [&](int kt) {
+        const int kk = kt * GEMM_BK;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            const int e = tid + i * GEMM_TPB;          // 0..2047
+            const int r = e >> 4, k = e & 15;          // row-major in tile
+            const int gr = row0 + r;
+            ra[i] = (gr < M && kk + k < K) ? A[(int64_t)gr * lda + kk + k] : 0.0;
+        }
+    };
+    auto load_b = [&](int kt) {
+        const int kk = kt * GEMM_BK;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            const int e = tid + i * GEMM_TPB;          // 0..2047
+            const int k = e >> 7, c = e & 127;
+            const int64_t gc = col0 + c;
+            rb[i] = (kk + k < K && gc < N) ? B[(int64_t)(kk + k) * ldb + gc] : 0.0;
+        }
+    };
+    auto write_lds = [&]() {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            const int e = tid + i * GEMM_TPB;
+            As[e & 15][e >> 4] = ra[i];
+        }
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            const int e = tid + i * GEMM_TPB;
+            Bs[e >> 7][e & 127] = rb[i];
+        }
+    };
+
+    load_a(0);
+    load_b(0);
+    write_lds();
+
+    for (int kt = 0; kt < ktiles; ++kt) {
+        __syncthreads();                 // LDS tile kt ready
+        if (kt + 1 < ktiles) {           // issue next tile's global loads
+            load_a(kt + 1);
+            load_b(kt + 1);
+        }
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            const int k = kk * 4 + fk;
+            double af[4], bf[4];
+#pragma unroll
+            for (int i = 0; i < 4; ++i) af[i] = As[k][wm0 + i * 16 + frow];
+#pragma unroll
+            for (int j = 0; j < 4; ++j) bf[j] = Bs[k][wn0 + j * 16 + frow];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();                 // everyone done reading tile kt
+        if (kt + 1 < ktiles) write_lds();
+    }
+
+    // epilogue: C -= acc   (read-modify-write, coalesced over frag columns)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int r = row0 + wm0 + i * 16 + fk * 4 + q;
+                const int64_t cidx = col0 + wn0 + j * 16 + frow;
+                if (r < M && cidx < N) {
+                    C[(int64_t)r * ldc + cidx] -= acc[i][j][q];
+                }
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// misc small kernels for the distributed path
+// ---------------------------------------------------------------------------
+// candidate pack: cand row i = [ gri[f+i] | A10[f+i, 0..v) ]  (prepend_column,
+// utils.hpp:13-26 + step-1 copy conflux_opt.hpp:698-705)
+// cand row i = [ gri[f+s] | A10[f+s, :] ] with s = idx ? idx[i] : i;
+// s >= n_src stands for the reference's zero-padded candidate rows
+// (step0_padding, conflux_opt.hpp:604-614).
+__global__ void k_pack_candidate(const double *__restrict__ A10, int64_t lda,
+                                 const int *__restrict__ gri, int f, int n_src,
+                                 int n_out, int v, const int *__restrict__ idx,
+                                 double *__restrict__ cand) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)n_out * (v + 1)) return;
+    const int r = (int)(i / (v + 1)), c = (int)(i % (v + 1));
+    const int s = idx ? idx[r] : r;
+    if (s >= n_src) { cand[i] = 0.0; return; }
+    cand[i] = (c == 0) ? (double)gri[f + s] : A10[(int64_t)(f + s) * lda + c - 1];
+}
+
+__global__ void k_extract_col0_int(const double *__restrict__ cand, int stride,
+                                   int n, int *__restrict__ out) {
+    const int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) out[i] = (int)cand[(int64_t)i * stride];
+}
+
+// slab pack for the A10 spread (conflux_opt.hpp:1389-1399): column slab
+// [pk*nlayr, (pk+1)*nlayr) of X (n x v) -> contiguous n x nlayr at slab pk
+__global__ void k_slab_pack(const double *__restrict__ X, int64_t ldx, int n,
+                            int nlayr, int Pz, double *__restrict__ out) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)n * nlayr * Pz) return;
+    const int pk = (int)(i / ((int64_t)n * nlayr));
+    const int64_t rest = i % ((int64_t)n * nlayr);
+    const int r = (int)(rest / nlayr), c = (int)(rest % nlayr);
+    out[i] = X[(int64_t)r * ldx + pk * nlayr + c];
+}
+
+}  // namespace ck
+
+// ---------------------------------------------------------------------------
+// launch wrappers (host side)
+// ---------------------------------------------------------------------------
+using namespace ck;
+
+static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
+
+void launch_init_matrix(double *A, int Ml, int Nl, int v, int Px, int Py,
+                        int pi, int pj, int zero_layer, uint64_t seed,
+                        hipStream_t s) {
+    const int64_t total = (int64_t)Ml * Nl;
+    hipLaunchKernelGGL(k_init_matrix, dim3(cdiv64(total, 256)), dim3(256), 0, s,
+                       A, Ml, Nl, v, Px, Py, pi, pj, zero_layer, seed);
+}
+
+void launch_copy2d(const double *src, int64_t lds, double *dst, int64_t ldd,
+                   int rows, int64_t cols, hipStream_t s) {
+    if (rows <= 0 || cols <= 0) return;
+    hipLaunchKernelGGL(k_copy2d, dim3(cdiv64((int64_t)rows * cols, 256)),
+                       dim3(256), 0, s, src, lds, dst, ldd, rows, cols);
+}
+
+void launch_zero2d(double *dst, int64_t ldd, int rows, int64_t cols,
+                   hipStream_t s) {
+    if (rows <= 0 || cols <= 0) return;
+    hipLaunchKernelGGL(k_zero2d, dim3(cdiv64((int64_t)rows * cols, 256)),
+                       dim3(256), 0, s, dst, ldd, rows, cols);
+}
+
+void launch_add2d(const double *src, int64_t lds, double *dst, int64_t ldd,
+                  int rows, int64_t cols, hipStream_t s) {
+    if (rows <= 0 || cols <= 0) return;
+    hipLaunchKernelGGL(k_add2d, dim3(cdiv64((int64_t)rows * cols, 256)),
+                       dim3(256), 0, s, src, lds, dst, ldd, rows, cols);
+}
+
+void launch_row_gather(const double *src, int64_t lds, double *dst,
+                       int64_t ldd, const int *idx, int n_rows, int64_t cols,
+                       hipStream_t s) {
+    if (n_rows <= 0 || cols <= 0) return;
+    hipLaunchKernelGGL(k_row_gather, dim3(cdiv64((int64_t)n_rows * cols, 256)),
+                       dim3(256), 0, s, src, lds, dst, ldd, idx, n_rows, cols);
+}
+
+void launch_row_scatter(const double *src, int64_t lds, double *dst,
+                        int64_t ldd, const int *idx, int n_rows, int64_t cols,
+                        hipStream_t s) {
+    if (n_rows <= 0 || cols <= 0) return;
+    hipLaunchKernelGGL(k_row_scatter, dim3(cdiv64((int64_t)n_rows * cols, 256)),
+                       dim3(256), 0, s, src, lds, dst, ldd, idx, n_rows, cols);
+}
+
+void launch_laswp(double *A, int64_t lda, int i0, const int *piv, int nswap,
+                  int64_t c0, int64_t c1, hipStream_t s) {
+    if (nswap <= 0 || c1 <= c0) return;
+    hipLaunchKernelGGL(k_laswp, dim3(cdiv64(c1 - c0, 256)), dim3(256), 0, s, A,
+                       lda, i0, piv, nswap, c0, c1);
+}
+
+void launch_cm_import(const double *src, int64_t lds, double *cm, int64_t ldc,
+                      int rows, int cols, hipStream_t s) {
+    if (rows <= 0 || cols <= 0) return;
+    hipLaunchKernelGGL(k_cm_import, dim3(cdiv64((int64_t)rows * cols, 256)),
+                       dim3(256), 0, s, src, lds, cm, ldc, rows, cols);
+}
+
+void launch_cm_export(const double *cm, int64_t ldc, double *dst, int64_t ldd,
+                      int rows, int cols, hipStream_t s) {
+    if (rows <= 0 || cols <= 0) return;
+    hipLaunchKernelGGL(k_cm_export, dim3(cdiv64((int64_t)rows * cols, 256)),
+                       dim3(256), 0, s, cm, ldc, dst, ldd, rows, cols);
+}
+
+void launch_panel_col(double *cm, int64_t ldc, int m, int nb, int c,
+                      void *sync, int *ipiv, unsigned int epoch,
+                      hipStream_t s) {
+    const int rows = m - c;
+    int nblocks = (int)cdiv64(rows, PANEL_RPB);
+    if (nblocks < 1) nblocks = 1;
+    if (nblocks > CONFLUX_PANEL_MAX_BLOCKS) nblocks = CONFLUX_PANEL_MAX_BLOCKS;
+    hipLaunchKernelGGL(k_panel_col, dim3(nblocks), dim3(PANEL_TPB), 0, s, cm,
+                       ldc, m, nb, c, (PanelSync *)sync, ipiv, epoch, nblocks);
+}
+
+int conflux_panel_sync_bytes() { return (int)sizeof(PanelSync); }
+int conflux_panel_nb() { return PANEL_NB; }
+
+void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
+                                   int64_t ldx, int nb, int64_t N,
+                                   hipStream_t s) {
+    if (N <= 0 || nb <= 0) return;
+    hipLaunchKernelGGL(k_trsm_left_lower_unit32, dim3(cdiv64(N, 256)),
+                       dim3(256), 0, s, L, ldl, X, ldx, nb, N);
+}
+
+void launch_trsm_right_upper32(const double *U, int64_t ldu, double *X,
+                               int64_t ldx, int nb, int64_t M, hipStream_t s) {
+    if (M <= 0 || nb <= 0) return;
+    hipLaunchKernelGGL(k_trsm_right_upper32, dim3(cdiv64(M, 256)), dim3(256),
+                       0, s, U, ldu, X, ldx, nb, M);
+}
+
+void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
+                      int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
+                      int K, hipStream_t s) {
+    if (M <= 0 || N <= 0 || K <= 0) return;
+    const int ntm = (int)cdiv64(M, GEMM_BM);
+    const int ntn = (int)cdiv64(N, GEMM_BN);
+    hipLaunchKernelGGL(k_dgemm_f64, dim3(ntm * ntn), dim3(GEMM_TPB), 0, s, A,
+                       lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
+}
+
+void launch_pack_candidate(const double *A10, int64_t lda, const int *gri,
+                           int f, int n_src, int n_out, int v, const int *idx,
+                           double *cand, hipStream_t s) {
+    if (n_out <= 0) return;
+    hipLaunchKernelGGL(k_pack_candidate,
+                       dim3(cdiv64((int64_t)n_out * (v + 1), 256)), dim3(256),
+                       0, s, A10, lda, gri, f, n_src, n_out, v, idx, cand);
+}
+
+void launch_row_move(const double *src, int64_t lds, double *dst, int64_t ldd,
+                     const int *src_idx, const int *dst_idx, int n_rows,
+                     int64_t cols, hipStream_t s) {
+    if (n_rows <= 0 || cols <= 0) return;
+    hipLaunchKernelGGL(k_row_move, dim3(cdiv64((int64_t)n_rows * cols, 256)),
+                       dim3(256), 0, s, src, lds, dst, ldd, src_idx, dst_idx,
+                       n_rows, cols);
+}
+
+void launch_extract_col0_int(const double *cand, int stride, int n, int *out,
+                             hipStream_t s) {
+    if (n <= 0) return;
+    hipLaunchKernelGGL(k_extract_col0_int, dim3(cdiv64(n, 256)), dim3(256), 0,
+                       s, cand, stride, n, out);
+}
+
+void launch_slab_pack(const double *X, int64_t ldx, int n, int nlayr, int Pz,
+                      double *out, hipStream_t s) {
+    if (n <= 0) return;
+    hipLaunchKernelGGL(k_slab_pack,
+                       dim3(cdiv64((int64_t)n * nlayr * Pz, 256)), dim3(256),
+                       0, s, X, ldx, n, nlayr, Pz, out);
+}

@@ -1,0 +1,51 @@
+// Host-side launch API of the HIP kernels (kernels.hip).
+#pragma once
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+#define CONFLUX_PANEL_MAX_BLOCKS 256
+
+void launch_init_matrix(double *A, int Ml, int Nl, int v, int Px, int Py,
+                        int pi, int pj, int zero_layer, uint64_t seed,
+                        hipStream_t s);
+void launch_copy2d(const double *src, int64_t lds, double *dst, int64_t ldd,
+                   int rows, int64_t cols, hipStream_t s);
+void launch_zero2d(double *dst, int64_t ldd, int rows, int64_t cols,
+                   hipStream_t s);
+void launch_add2d(const double *src, int64_t lds, double *dst, int64_t ldd,
+                  int rows, int64_t cols, hipStream_t s);
+void launch_row_gather(const double *src, int64_t lds, double *dst,
+                       int64_t ldd, const int *idx, int n_rows, int64_t cols,
+                       hipStream_t s);
+void launch_row_scatter(const double *src, int64_t lds, double *dst,
+                        int64_t ldd, const int *idx, int n_rows, int64_t cols,
+                        hipStream_t s);
+void launch_laswp(double *A, int64_t lda, int i0, const int *piv, int nswap,
+                  int64_t c0, int64_t c1, hipStream_t s);
+void launch_cm_import(const double *src, int64_t lds, double *cm, int64_t ldc,
+                      int rows, int cols, hipStream_t s);
+void launch_cm_export(const double *cm, int64_t ldc, double *dst, int64_t ldd,
+                      int rows, int cols, hipStream_t s);
+void launch_panel_col(double *cm, int64_t ldc, int m, int nb, int c,
+                      void *sync, int *ipiv, unsigned int epoch, hipStream_t s);
+int conflux_panel_sync_bytes();
+int conflux_panel_nb();
+void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
+                                   int64_t ldx, int nb, int64_t N,
+                                   hipStream_t s);
+void launch_trsm_right_upper32(const double *U, int64_t ldu, double *X,
+                               int64_t ldx, int nb, int64_t M, hipStream_t s);
+void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
+                      int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
+                      int K, hipStream_t s);
+void launch_pack_candidate(const double *A10, int64_t lda, const int *gri,
+                           int f, int n_src, int n_out, int v, const int *idx,
+                           double *cand, hipStream_t s);
+void launch_row_move(const double *src, int64_t lds, double *dst, int64_t ldd,
+                     const int *src_idx, const int *dst_idx, int n_rows,
+                     int64_t cols, hipStream_t s);
+void launch_extract_col0_int(const double *cand, int stride, int n, int *out,
+                             hipStream_t s);
+void launch_slab_pack(const double *X, int64_t ldx, int n, int nlayr, int Pz,
+                      double *out, hipStream_t s);

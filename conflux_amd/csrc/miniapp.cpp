@@ -1,0 +1,179 @@
+// conflux_miniapp — drop-in CLI for the reference miniapp
+// (reference examples/conflux_miniapp.cpp:42-84 flags, :119/:156-165 output).
+//
+//   conflux_miniapp -N <n> -b <block> [--p_grid=Px,Py,Pz] [-r reps]
+//                   [-l print_limit] [-t type] [--sim] [--timing]
+//
+// Differences from the reference (documented in DESIGN.md):
+//   * multi-rank runs self-spawn one process per GPU (fork + pipes for the
+//     RCCL unique id) instead of requiring mpirun;
+//   * --sim runs all ranks of the grid in ONE process on ONE GPU
+//     (choreography-identical, D2D transport) — used for 1-GPU validation;
+//   * --timing disables factor collection (the reference's non-VALIDATION
+//     build); default keeps it on like CONFLUX_WITH_VALIDATION.
+#include <hip/hip_runtime.h>
+#include <sys/wait.h>
+#include <unistd.h>
+
+#include <cmath>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include "../../include/conflux_lu.h"
+
+static void usage() {
+    std::printf(
+        "conflux miniapp (MI355X engine)\n"
+        "  -N, --cols N          matrix dimension (default 1000 -> rounded)\n"
+        "  -b, --block_size b    tile size v (default 256)\n"
+        "  -p, --p_grid Px,Py,Pz process grid (default 1,1,1)\n"
+        "  -r, --n_rep r         repetitions (default 2)\n"
+        "  -l, --print_limit l   (accepted for compatibility)\n"
+        "  -t, --type t          weak|strong|other (label only)\n"
+        "      --sim             all ranks in one process on one GPU\n"
+        "      --timing          skip factor collection (bench mode)\n");
+}
+
+int main(int argc, char **argv) {
+    int N = 1000, b = 256, reps = 2, Px = 1, Py = 1, Pz = 1;
+    std::string type = "other";
+    bool sim = false, timing = false;
+    for (int i = 1; i < argc; ++i) {
+        std::string a = argv[i];
+        auto val = [&](const char *) -> std::string {
+            if (a.find('=') != std::string::npos) return a.substr(a.find('=') + 1);
+            return (i + 1 < argc) ? argv[++i] : "";
+        };
+        if (a == "-h" || a == "--help") { usage(); return 0; }
+        else if (a == "-N" || a.rfind("--cols", 0) == 0) N = std::atoi(val("N").c_str());
+        else if (a == "-b" || a.rfind("--block_size", 0) == 0) b = std::atoi(val("b").c_str());
+        else if (a == "-r" || a.rfind("--n_rep", 0) == 0) reps = std::atoi(val("r").c_str());
+        else if (a == "-l" || a.rfind("--print_limit", 0) == 0) (void)val("l");
+        else if (a == "-t" || a.rfind("--type", 0) == 0) type = val("t");
+        else if (a == "--sim") sim = true;
+        else if (a == "--timing") timing = true;
+        else if (a == "-p" || a.rfind("--p_grid", 0) == 0) {
+            std::string g = val("p");
+            if (std::sscanf(g.c_str(), "%d,%d,%d", &Px, &Py, &Pz) != 3) {
+                std::fprintf(stderr, "bad --p_grid\n");
+                return 1;
+            }
+        } else {
+            std::fprintf(stderr, "unknown arg %s\n", a.c_str());
+            return 1;
+        }
+    }
+    if (Px <= 0 || Py <= 0 || Pz <= 0) { Px = Py = Pz = 1; }
+    const int P = Px * Py * Pz;
+    // round N like the reference (lu_params.hpp:67-71)
+    const int ntx = (N + b * Px - 1) / (b * Px);
+    N = b * Px * ntx;
+
+    // launch topology: rank from env (external launcher), else self-spawn
+    int rank = -2, world = P;
+    char uid[CONFLUX_LU_UID_BYTES];
+    std::vector<int> kids;
+    if (sim || P == 1) {
+        rank = sim ? -1 : 0;
+        world = P;
+    } else if (const char *er = std::getenv("CONFLUX_RANK")) {
+        rank = std::atoi(er);
+        // uid passed via fd 3 is only for self-spawn; external launchers use
+        // CONFLUX_UID_FILE
+        if (const char *uf = std::getenv("CONFLUX_UID_FILE")) {
+            FILE *f = std::fopen(uf, "rb");
+            if (!f || std::fread(uid, 1, sizeof uid, f) != sizeof uid) {
+                std::fprintf(stderr, "cannot read uid file\n");
+                return 1;
+            }
+            std::fclose(f);
+        }
+    }
+    // (rank == -2 falls through to the self-spawn path below)
+    (void)kids;
+
+    if (rank == -2) {
+        // self-spawn path
+        int ndev = 0;
+        (void)hipGetDeviceCount(&ndev);
+        if (ndev < P) {
+            std::fprintf(stderr,
+                         "[conflux_miniapp] %d GPUs visible but grid needs %d "
+                         "(use --sim for single-GPU validation)\n",
+                         ndev, P);
+            return 1;
+        }
+        std::vector<int> pipes(2 * P);
+        for (int r = 0; r < P; ++r)
+            if (pipe(&pipes[2 * r])) { perror("pipe"); return 1; }
+        std::vector<pid_t> pids(P);
+        for (int r = 0; r < P; ++r) {
+            pid_t pid = fork();
+            if (pid == 0) {
+                (void)hipSetDevice(r);
+                if (r == 0) {
+                    if (conflux_lu_make_uid(uid)) return 1;
+                    for (int q = 1; q < P; ++q)
+                        if (write(pipes[2 * q + 1], uid, sizeof uid) !=
+                            (ssize_t)sizeof uid)
+                            return 1;
+                } else {
+                    if (read(pipes[2 * r], uid, sizeof uid) !=
+                        (ssize_t)sizeof uid)
+                        return 1;
+                }
+                rank = r;
+                goto run;
+            }
+            pids[r] = pid;
+        }
+        {
+            int status = 0, bad = 0;
+            for (int r = 0; r < P; ++r) {
+                waitpid(pids[r], &status, 0);
+                if (!WIFEXITED(status) || WEXITSTATUS(status)) bad = 1;
+            }
+            return bad;
+        }
+    }
+
+run:
+    conflux_lu_ctx *ctx = nullptr;
+    int rc = conflux_lu_create(N, b, Px, Py, Pz, rank, world,
+                               (world > 1) ? uid : nullptr, &ctx);
+    if (rc) {
+        std::fprintf(stderr, "[conflux_miniapp] create failed rc=%d\n", rc);
+        return 1;
+    }
+    conflux_lu_store_factors(ctx, timing ? 0 : 1);
+
+    const bool print0 = (rank <= 0);
+    if (print0) {
+        int Ml, Nl, Nt, nlayr, M, Np;
+        conflux_lu_dims(ctx, &Ml, &Nl, &Nt, &nlayr, &M, &Np);
+        std::printf("======== INTERNAL PARAMS ========\n");
+        std::printf("M: %d, N: %d, P: %d, v: %d, Px: %d, Py: %d, Pz: %d, Nt: %d\n",
+                    M, Np, P, b, Px, Py, Pz, Nt);
+        std::printf("======== RESULT FORMAT ========\n");
+        std::printf("_result_ lu,conflux,<num_rows>,<num_cols>,<num_ranks>,"
+                    "<process_grid>,time,other,<time_in_ms>,<block_size>\n");
+        std::printf("======== RESULTS ========\n");
+    }
+
+    const int sqrtP = (int)std::max(1.0, std::floor(std::sqrt((double)P)));
+    const int N_base = (type == "weak") ? N / sqrtP : N;
+    for (int i = 0; i < reps + 1; ++i) {
+        conflux_lu_init_matrix(ctx, 42);
+        double ms = 0;
+        rc = conflux_lu_factor(ctx, &ms);
+        if (rc) { std::fprintf(stderr, "factor failed rc=%d\n", rc); return 1; }
+        if (i > 0 && print0)
+            std::printf("_result_ lu,conflux,%d,%d,%d,%dx%dx%d,time,%s,%.0f,%d\n",
+                        N, N_base, P, Px, Py, Pz, type.c_str(), ms, b);
+    }
+    conflux_lu_destroy(ctx);
+    return 0;
+}

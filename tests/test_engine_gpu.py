@@ -1,0 +1,173 @@
+"""GPU parity tests: the HIP engine against the oracle (which is itself
+pinned to the compiled reference — tests/test_oracle.py).
+
+All tests run on ONE MI355X.  Multi-rank grids run in the engine's
+single-process simulation mode: identical choreography and kernels as the
+RCCL path, device-to-device transport.
+"""
+import ctypes
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+from oracle import Params, gen_matrix, lu_oracle, residual_check  # noqa: E402
+
+TOL_F = 1e-11
+TOL_RES = 1e-14
+
+
+@pytest.fixture(scope="module")
+def eng():
+    import conflux_amd
+    return conflux_amd
+
+
+# ---------------- kernel-level numerics (vs numpy fp64 reference) ----------
+
+def test_dgemm_numerics(eng):
+    rng = np.random.default_rng(0)
+    for (M, N, K) in [(128, 128, 16), (256, 384, 32), (130, 257, 48),
+                      (1, 128, 16), (300, 1, 512), (512, 512, 512)]:
+        A = rng.standard_normal((M, K))
+        B = rng.standard_normal((K, N))
+        C = rng.standard_normal((M, N))
+        C1 = C.copy()
+        rc = eng.lib().conflux_lu_debug_dgemm(
+            M, N, K, A.ctypes.data_as(ctypes.c_void_p),
+            B.ctypes.data_as(ctypes.c_void_p),
+            C1.ctypes.data_as(ctypes.c_void_p))
+        assert rc == 0
+        ref = C - A @ B
+        err = np.abs(C1 - ref).max() / (np.abs(ref).max() + 1)
+        assert err < 1e-13, f"dgemm M={M} N={N} K={K}: err={err}"
+
+
+def test_dgemm_transpose_detecting(eng):
+    # asymmetric A and B catch any row/col swap in the MFMA C layout (G9)
+    M = N = 64
+    K = 16
+    A = np.arange(M * K, dtype=np.float64).reshape(M, K) / 100
+    B = (np.arange(K * N, dtype=np.float64).reshape(K, N) ** 1.5) / 1000
+    C = np.zeros((M, N))
+    C1 = C.copy()
+    assert eng.lib().conflux_lu_debug_dgemm(
+        M, N, K, A.ctypes.data_as(ctypes.c_void_p),
+        B.ctypes.data_as(ctypes.c_void_p),
+        C1.ctypes.data_as(ctypes.c_void_p)) == 0
+    assert np.allclose(C1, -A @ B, atol=1e-10)
+
+
+def test_getrf_matches_lapack(eng):
+    import scipy.linalg as la
+    rng = np.random.default_rng(1)
+    for (n, v) in [(64, 32), (96, 32), (128, 64), (257, 64), (1024, 128)]:
+        P0 = 5 + rng.random((n, v))
+        P1 = np.ascontiguousarray(P0)
+        ipiv = np.zeros(v, dtype=np.int32)
+        rc = eng.lib().conflux_lu_debug_getrf(
+            n, v, P1.ctypes.data_as(ctypes.c_void_p),
+            ipiv.ctypes.data_as(ctypes.c_void_p))
+        assert rc == 0
+        lu, piv, info = la.lapack.dgetrf(np.asfortranarray(P0))
+        nst = min(n, v)
+        assert np.array_equal(ipiv[:nst], piv[:nst]), f"pivots n={n} v={v}"
+        assert np.abs(P1 - lu).max() < 1e-12, f"factors n={n} v={v}"
+
+
+def test_trsm_right_upper(eng):
+    rng = np.random.default_rng(2)
+    v, M = 64, 500
+    U = np.triu(rng.random((v, v)) + np.eye(v) * 5)
+    X = rng.standard_normal((M, v))
+    X1 = X.copy()
+    assert eng.lib().conflux_lu_debug_trsm(
+        1, M, M, v, U.ctypes.data_as(ctypes.c_void_p),
+        X1.ctypes.data_as(ctypes.c_void_p)) == 0
+    ref = np.linalg.solve(U.T, X.T).T
+    assert np.abs(X1 - ref).max() < 1e-12
+
+
+def test_trsm_left_lower_unit(eng):
+    rng = np.random.default_rng(3)
+    v, N = 64, 700
+    L = np.tril(rng.random((v, v)), -1) + np.eye(v)
+    X = rng.standard_normal((v, N))
+    X1 = X.copy()
+    assert eng.lib().conflux_lu_debug_trsm(
+        0, v, N, v, L.ctypes.data_as(ctypes.c_void_p),
+        X1.ctypes.data_as(ctypes.c_void_p)) == 0
+    ref = np.linalg.solve(L, X)
+    assert np.abs(X1 - ref).max() < 1e-12
+
+
+# ---------------- end-to-end parity vs the oracle ---------------------------
+
+GRIDS = [
+    (64, 8, 1, 1, 1),
+    (128, 16, 1, 1, 1),
+    (128, 16, 1, 1, 2),
+    (64, 8, 2, 2, 1),
+    (64, 8, 2, 2, 2),
+    (128, 16, 2, 2, 2),
+    (128, 8, 4, 4, 2),
+    (256, 32, 2, 2, 1),
+]
+
+
+@pytest.mark.parametrize("N,v,Px,Py,Pz", GRIDS)
+def test_lu_parity(eng, N, v, Px, Py, Pz):
+    A = gen_matrix(N)
+    p = Params(N, v, Px, Py, Pz)
+    r = lu_oracle(A, p)
+    with eng.Engine(N, v, Px, Py, Pz, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor()
+        perm = e.get_perm()
+        F = e.get_F_global()
+    assert np.array_equal(perm, r["perm"]), "pivot indices must be bit-exact"
+    assert np.abs(F - r["F"]).max() < TOL_F
+    assert residual_check(A, perm, F) < TOL_RES
+
+
+def test_lu_parity_golden(eng, golden):
+    """Engine vs the compiled reference's own outputs (golden fixtures)."""
+    checked = 0
+    for tag in sorted({k.split("/")[0] for k in golden.files}):
+        N, v, Px, Py, Pz = (int(x) for x in golden[f"{tag}/cfg"])
+        if N > 256:
+            continue
+        A = golden[f"{tag}/A"]
+        with eng.Engine(N, v, Px, Py, Pz, rank=-1) as e:
+            e.store_factors(True)
+            e.set_matrix_global(A)
+            e.factor()
+            perm = e.get_perm()
+            F = e.get_F_global()
+        assert np.array_equal(perm, golden[f"{tag}/perm"]), f"{tag} pivots"
+        assert np.abs(F - golden[f"{tag}/C"]).max() < TOL_F, f"{tag} factors"
+        checked += 1
+    assert checked >= 8
+
+
+@pytest.mark.parametrize("N,v", [(1024, 128), (2048, 256)])
+def test_lu_single_rank_medium(eng, N, v):
+    """Size-scaling property check at sizes the oracle still runs fast."""
+    A = gen_matrix(N)
+    with eng.Engine(N, v, 1, 1, 1, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor()
+        perm = e.get_perm()
+        F = e.get_F_global()
+    # size-independent property: ||PA - LU||/||A|| at fp64
+    assert residual_check(A, perm, F) < 1e-13
+    # pivots vs LAPACK partial pivoting (1x1x1 degenerates to getrf)
+    import scipy.linalg as la
+    lu, piv = la.lu_factor(A)
+    ref = np.arange(N)
+    for i, j in enumerate(piv):
+        ref[i], ref[j] = ref[j], ref[i]
+    assert np.array_equal(perm, ref)
