@@ -488,6 +488,9 @@ int run_step(Ctx &c, int k) {
             launch_copy2d(r.panel, v, r.A00, v, std::min(v, n), v, c.stream);
             for (int i = 0; i < v; ++i)
                 sp.gpivots[i] = (perm[i] < n) ? r.gri[r.fnp + perm[i]] : 0;
+            if (!c.sim && c.world > 1)  // feed the C4 broadcast below
+                HIPCHK(hipMemcpyAsync(r.d_gpivots, sp.gpivots.data(), v * 4,
+                                      hipMemcpyHostToDevice, c.stream));
         } else {
             // winners = candidate rows perm[:v] (id col + A10 row), placed
             // top/bottom by the round-0 pairing (conflux_opt.hpp:724,741-751)
@@ -554,12 +557,14 @@ int run_step(Ctx &c, int k) {
         }
     }
 
-    if (n_rounds > 0) {
+    if (n_rounds > 0 || (!c.sim && c.world > 1)) {
         // extract gpivots from candidate col 0 (conflux_opt.hpp:810-816)
-        for (auto &r : c.rs) {
-            if (r.pj != kcol || r.pk != 0) continue;
-            launch_extract_col0_int(r.cand, v + 1, v, r.d_gpivots, c.stream);
-        }
+        if (n_rounds > 0)
+            for (auto &r : c.rs) {
+                if (r.pj != kcol || r.pk != 0) continue;
+                launch_extract_col0_int(r.cand, v + 1, v, r.d_gpivots,
+                                        c.stream);
+            }
         // A00 transpose-pair exchange (C3) + gpivots broadcast (C4)
         if (!c.sim) {
             RankState &me = c.rs[0];
@@ -608,9 +613,6 @@ int run_step(Ctx &c, int k) {
                 }
             HIPCHK(hipStreamSynchronize(c.stream));
         }
-    } else if (!c.sim && c.world > 1) {
-        c.err = "Px==1 with world>1 unsupported";
-        return CONFLUX_LU_EARG;
     }
 
     plan_from_gpivots(c, sp);
@@ -897,8 +899,16 @@ int run_step(Ctx &c, int k) {
             return CONFLUX_LU_EINTERNAL;
         if (c.store_factors) {
             const int ltik = k / Px;
-            launch_copy2d(r.A01, wA01, r.Fres + i64(ltik) * v * Nl + loff, Nl,
-                          v, wA01, c.stream);
+            // U region starts at this rank's first local column tile with
+            // global tile >= k; columns left of it carry stale
+            // already-factored data the reference never reads
+            // (cf. oracle lu_oracle.py step-5 `gcs >= off` mask)
+            const int64_t ustart =
+                i64(v) * (r.pj < kcol ? k / Py + 1 : k / Py);
+            if (Nl - ustart > 0)
+                launch_copy2d(r.A01 + (ustart - loff), wA01,
+                              r.Fres + i64(ltik) * v * Nl + ustart, Nl, v,
+                              Nl - ustart, c.stream);
             if (r.pj == kcol)  // diagonal tile: packed LU from A00
                 launch_copy2d(r.A00, v, r.Fres + i64(ltik) * v * Nl + loff, Nl,
                               v, v, c.stream);

@@ -550,7 +550,9 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
         for (int j = 0; j < 4; ++j) {
 #pragma unroll
             for (int q = 0; q < 4; ++q) {
-                const int r = row0 + wm0 + i * 16 + fk * 4 + q;
+                // f64 16x16x4 C/D map (hardware-verified by mfma_probe):
+                // lane l, reg q -> D[4*q + (l>>4)][l & 15]
+                const int r = row0 + wm0 + i * 16 + q * 4 + fk;
                 const int64_t cidx = col0 + wn0 + j * 16 + frow;
                 if (r < M && cidx < N) {
                     C[(int64_t)r * ldc + cidx] -= acc[i][j][q];

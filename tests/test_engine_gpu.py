@@ -64,7 +64,7 @@ def test_getrf_matches_lapack(eng):
     rng = np.random.default_rng(1)
     for (n, v) in [(64, 32), (96, 32), (128, 64), (257, 64), (1024, 128)]:
         P0 = 5 + rng.random((n, v))
-        P1 = np.ascontiguousarray(P0)
+        P1 = P0.copy()  # debug_getrf factors in place — keep P0 pristine
         ipiv = np.zeros(v, dtype=np.int32)
         rc = eng.lib().conflux_lu_debug_getrf(
             n, v, P1.ctypes.data_as(ctypes.c_void_p),
