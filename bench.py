@@ -1,0 +1,203 @@
+#!/usr/bin/env python3
+"""bench.py — headline benchmark of the MI355X-native CONFLUX LU engine.
+
+Measures BASELINE.json's metric — fp64 LU TFLOP/s, (2/3)N^3 / t_wall — on the
+BASELINE configs.  A "step" is one full LU factorization of the synthetic
+N x N fp64 matrix (the reference miniapp's repetition unit,
+examples/conflux_miniapp.cpp:138-167).
+
+    python bench.py [--gpus N] [--steps K] [--warmup W]
+
+Single process when N==1; for N>1 the driver launches this file under
+torch.distributed.run with one rank per GPU — torch.distributed (gloo) is
+used ONLY to broadcast the RCCL unique id and for host barriers; all compute
+and all data-path communication happen inside libconflux_lu.so (HIP + RCCL
+over xGMI).
+
+Workloads (grid restrictions Px==Py pow2):
+    1 GPU : N=16384, v=512, grid 1x1x1   (BASELINE config 2)
+    2 GPU : N=16384, v=512, grid 1x1x2   (depth replication)
+    4 GPU : N=32768, v=512, grid 2x2x1   (BASELINE config 3)
+    8 GPU : N=65536, v=512, grid 2x2x2   (BASELINE config 4)
+
+Rank 0 prints ONE JSON line per the driver contract.
+"""
+import argparse
+import json
+import os
+import subprocess
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+GRIDS = {
+    1: (16384, 512, 1, 1, 1),
+    2: (16384, 512, 1, 1, 2),
+    4: (32768, 512, 2, 2, 1),
+    8: (65536, 512, 2, 2, 2),
+}
+
+FP64_MFMA_PEAK_TFLOPS = 78.6  # gfx950: 256 CU x 2.4 GHz x 128 f64 flop/clk/CU
+
+
+def cpu_baseline():
+    """Time the compiled reference (oracle/_ref, MKL+MPICH — kind
+    'reference') on this box's host cores.  Bounded sample: N=8192, v=512,
+    grid 2x2x1 under mpiexec -n 4 (the reference cannot run 1-wide grids),
+    one repetition (~10-30 s of CPU work).  Reported baseline, not target."""
+    ref = os.path.join(os.path.dirname(os.path.abspath(__file__)), "oracle",
+                       "_ref", "conflux_ref")
+    if not os.path.exists(ref):
+        return None
+    ncores = os.cpu_count() or 1
+    n_ranks = 4
+    omp = max(1, ncores // n_ranks)
+    env = dict(os.environ, MKL_THREADING_LAYER="GNU",
+               OMP_NUM_THREADS=str(omp), LD_LIBRARY_PATH="/opt/conda/lib")
+    N, v = 8192, 512
+    try:
+        out = subprocess.run(
+            ["/opt/conda/bin/mpiexec", "-n", str(n_ranks), ref, str(N),
+             str(v), "2", "2", "1", "-", "/tmp/confluxref_bench", "1"],
+            env=env, capture_output=True, text=True, timeout=900)
+        ms = None
+        for line in out.stdout.splitlines():
+            if line.startswith("_result_"):
+                ms = float(line.split(",")[-2])
+        if ms is None:
+            return None
+        tflops = (2.0 / 3.0) * N ** 3 / (ms * 1e-3) / 1e12
+        return {"value": round(tflops, 4), "unit": "TFLOP/s",
+                "cores": n_ranks * omp, "kind": "reference",
+                "sample": f"reference CPU path (MKL+MPICH) N={N} v={v} "
+                          f"grid 2x2x1, {n_ranks} ranks x {omp} OMP threads, "
+                          f"1 rep on this box's host cores"}
+    except Exception:
+        return None
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=None)
+    ap.add_argument("--steps", type=int, default=2)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--N", type=int, default=None, help="override matrix dim")
+    ap.add_argument("--v", type=int, default=None, help="override tile size")
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = args.gpus or world
+
+    # pin this process to its GPU BEFORE the HIP runtime loads
+    if world > 1:
+        os.environ["HIP_VISIBLE_DEVICES"] = str(local_rank)
+
+    N, v, Px, Py, Pz = GRIDS[n_gpus]
+    if args.N:
+        N = args.N
+    if args.v:
+        v = args.v
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        tdist.init_process_group("gloo", rank=rank, world_size=world)
+        dist = tdist
+
+    from conflux_amd import Engine
+
+    uid = None
+    if world > 1:
+        objs = [Engine.make_uid()] if rank == 0 else [None]
+        dist.broadcast_object_list(objs, src=0)
+        uid = objs[0]
+
+    eng = Engine(N, v, Px, Py, Pz, rank=(0 if world == 1 else rank),
+                 world=world, uid=uid)
+    eng.store_factors(False)  # timing mode (factor collection off, like the
+    #                           reference's non-VALIDATION build)
+
+    def barrier():
+        if dist:
+            dist.barrier()
+
+    flops_per_step = (2.0 / 3.0) * N ** 3
+
+    for _ in range(args.warmup):
+        eng.init_matrix(42)
+        eng.factor()
+    barrier()
+    t0 = time.perf_counter()
+    engine_ms = []
+    for _ in range(args.steps):
+        eng.init_matrix(42)
+        engine_ms.append(eng.factor())  # factor() ends with a device sync
+    barrier()
+    t1 = time.perf_counter()
+
+    total_s = t1 - t0
+    if dist:
+        import torch
+        tt = torch.tensor([total_s])
+        dist.all_reduce(tt, op=dist.ReduceOp.MAX)
+        total_s = float(tt[0])
+
+    stats = eng.kernel_stats()
+    g = stats["dgemm_trailing"]
+    roofline = None
+    if g["launches"] > 0 and g["seconds"] > 0:
+        achieved = g["flops"] / g["seconds"] / 1e12
+        roofline = {
+            "bound": "mfma",
+            "achieved": round(achieved, 3),
+            "peak": FP64_MFMA_PEAK_TFLOPS,
+            "unit": "TFLOP/s",
+            "frac": round(achieved / FP64_MFMA_PEAK_TFLOPS, 4),
+            "traffic": None,
+        }
+
+    eng.close()
+
+    if rank == 0:
+        tflops = args.steps * flops_per_step / total_s / 1e12
+        result = {
+            "metric": "fp64_lu_tflops",
+            "value": round(tflops, 3),
+            "unit": "TFLOP/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(total_s * 1e3 / args.steps, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": f"conflux LU N={N} v={v} grid {Px}x{Py}x{Pz} "
+                            f"(BASELINE cfg for {n_gpus} GPU)",
+                "N": N, "v": v, "grid": f"{Px}x{Py}x{Pz}",
+            },
+            "engine_ms_per_step": [round(x, 1) for x in engine_ms],
+            "kernel_stats": {k: {"seconds": round(s["seconds"], 4),
+                                 "launches": s["launches"]}
+                             for k, s in stats.items()},
+        }
+        if roofline:
+            result["roofline"] = roofline
+        if n_gpus == 1 and not args.skip_cpu_baseline:
+            cb = cpu_baseline()
+            if cb:
+                result["cpu_baseline"] = cb
+        print(json.dumps(result))
+
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -51,6 +51,9 @@ def _load():
         ctypes.c_void_p, ctypes.c_void_p]
     lib.conflux_lu_debug_getrf.argtypes = [ctypes.c_int, ctypes.c_int,
                                            ctypes.c_void_p, ctypes.c_void_p]
+    lib.conflux_lu_debug_dgemm_bench.argtypes = [
+        ctypes.c_int, ctypes.c_longlong, ctypes.c_int, ctypes.c_int,
+        ctypes.POINTER(ctypes.c_double)]
     lib.conflux_lu_debug_trsm.argtypes = [
         ctypes.c_int, ctypes.c_int, ctypes.c_longlong, ctypes.c_int,
         ctypes.c_void_p, ctypes.c_void_p]

@@ -135,7 +135,6 @@ int alloc_rank(Ctx &c, RankState &r, int pi, int pj, int pk) {
     HIPCHK(hipMalloc(&r.A00, v * v * 8));
     HIPCHK(hipMalloc(&r.cand, i64(2 * c.v) * (v + 1) * 8));
     HIPCHK(hipMalloc(&r.panel, prows * v * 8));
-    HIPCHK(hipMalloc(&r.cm, Ml * i64(conflux_panel_nb()) * 8));
     HIPCHK(hipMalloc(&r.A01pack, v * Nl * 8));
     HIPCHK(hipMalloc(&r.rowtmp, v * Nl * 8));
     HIPCHK(hipMalloc(&r.redtmp, i64(std::max(1, c.Pz - 1)) * Ml * v * 8));
@@ -203,18 +202,15 @@ int factor_panel(Ctx &c, RankState &r, int n, std::vector<int> &ipiv_out) {
     ipiv_out.assign(v, 0);
     size_t slot;
     if (ev_begin(c, 1, 0, &slot)) return CONFLUX_LU_EHIP;
-    std::vector<int> ipiv_sub(NB);
     for (int jb = 0; jb < nsteps; jb += NB) {
         const int nb = std::min(NB, nsteps - jb);
         const int m = n - jb;  // rows of the sub-panel
-        launch_cm_import(r.panel + i64(jb) * v + jb, v, r.cm, m, m, nb,
-                         c.stream);
-        for (int col = 0; col < nb; ++col) {
-            launch_panel_col(r.cm, m, m, nb, col, r.sync, r.d_ipiv + jb + col,
-                             c.epoch++, c.stream);
+        if (launch_panel_factor(r.panel + i64(jb) * v + jb, v, m, nb, r.sync,
+                                r.d_ipiv + jb, c.epoch, c.stream)) {
+            c.err = "panel grid not resident";
+            return CONFLUX_LU_EINTERNAL;
         }
-        launch_cm_export(r.cm, m, r.panel + i64(jb) * v + jb, v, m, nb,
-                         c.stream);
+        c.epoch += nb;
         // apply the sub-panel's swaps to the rest of the panel width
         launch_laswp(r.panel, v, jb, r.d_ipiv + jb, nb, 0, jb, c.stream);
         launch_laswp(r.panel, v, jb, r.d_ipiv + jb, nb, jb + nb, v, c.stream);
@@ -1180,13 +1176,41 @@ int conflux_lu_debug_dgemm(int M, int64_t N, int K, const double *A,
     return CONFLUX_LU_OK;
 }
 
+int conflux_lu_debug_dgemm_bench(int M, int64_t N, int K, int iters,
+                                 double *tflops) {
+    hipStream_t s;
+    HIPCHK(hipStreamCreate(&s));
+    double *dA, *dB, *dC;
+    HIPCHK(hipMalloc(&dA, i64(M) * K * 8));
+    HIPCHK(hipMalloc(&dB, i64(K) * N * 8));
+    HIPCHK(hipMalloc(&dC, i64(M) * N * 8));
+    launch_init_matrix(dA, M, K, K, 1, 1, 0, 0, 0, 7, s);
+    launch_init_matrix(dB, K, (int)N, (int)N, 1, 1, 0, 0, 0, 8, s);
+    launch_init_matrix(dC, M, (int)N, (int)N, 1, 1, 0, 0, 0, 9, s);
+    launch_dgemm_f64(dA, K, dB, N, dC, N, M, N, K, s);  // warmup
+    hipEvent_t a, b;
+    HIPCHK(hipEventCreate(&a));
+    HIPCHK(hipEventCreate(&b));
+    HIPCHK(hipEventRecord(a, s));
+    for (int i = 0; i < iters; ++i)
+        launch_dgemm_f64(dA, K, dB, N, dC, N, M, N, K, s);
+    HIPCHK(hipEventRecord(b, s));
+    HIPCHK(hipEventSynchronize(b));
+    float ms = 0;
+    HIPCHK(hipEventElapsedTime(&ms, a, b));
+    *tflops = 2.0 * M * (double)N * K * iters / (ms * 1e-3) / 1e12;
+    (void)hipFree(dA); (void)hipFree(dB); (void)hipFree(dC);
+    (void)hipEventDestroy(a); (void)hipEventDestroy(b);
+    (void)hipStreamDestroy(s);
+    return CONFLUX_LU_OK;
+}
+
 int conflux_lu_debug_getrf(int n, int v, double *panel, int *ipiv_out) {
     Ctx c;
     c.v = v;
     HIPCHK(hipStreamCreate(&c.stream));
     RankState r;
     HIPCHK(hipMalloc(&r.panel, i64(std::max(n, 2 * v)) * v * 8));
-    HIPCHK(hipMalloc(&r.cm, i64(n) * conflux_panel_nb() * 8));
     HIPCHK(hipMalloc(&r.d_ipiv, (v + 8) * 4));
     HIPCHK(hipMalloc(&r.sync, conflux_panel_sync_bytes()));
     HIPCHK(hipMemset(r.sync, 0, conflux_panel_sync_bytes()));
@@ -1198,7 +1222,7 @@ int conflux_lu_debug_getrf(int n, int v, double *panel, int *ipiv_out) {
                          hipMemcpyDeviceToHost));
         std::copy(ipiv.begin(), ipiv.end(), ipiv_out);
     }
-    (void)hipFree(r.panel); (void)hipFree(r.cm); (void)hipFree(r.d_ipiv);
+    (void)hipFree(r.panel); (void)hipFree(r.d_ipiv);
     (void)hipFree(r.sync);
     for (auto &e : c.evs) { (void)hipEventDestroy(e.a); (void)hipEventDestroy(e.b); }
     (void)hipStreamDestroy(c.stream);

@@ -23,6 +23,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
+#include <cstdlib>
 
 #include "kernels.hpp"
 
@@ -185,19 +186,29 @@ __global__ void k_cm_export(const double *__restrict__ cm, int64_t ldc,
 #define PANEL_RPB 512           // rows per block (2 per thread)
 #define PANEL_NB 32             // sub-panel width == register column budget
 
-struct PanelSync {
-    // per-block candidate: |value| bits, row, flag(epoch)
-    unsigned long long cand_abs[CONFLUX_PANEL_MAX_BLOCKS];
-    int cand_row[CONFLUX_PANEL_MAX_BLOCKS];
-    unsigned int cand_flag[CONFLUX_PANEL_MAX_BLOCKS];
-    // publication: new row c content (= old pivot row) and new piv row
-    // content (= old row c), columns c..nb-1 at slot index (col - c)
-    double rowc[PANEL_NB];
-    double rowpiv[PANEL_NB];
-    int piv_row;
+// --------------------------------------------------------------------------
+// persistent sub-panel factorization (v2): ONE launch factors a whole
+// nb(<=32)-wide sub-panel.  Each block owns PANEL_RPB rows, held in LDS for
+// all nb columns; the only cross-block traffic per column is the slab
+// handshake: every block publishes its candidate row (full nb values) and
+// the block owning the diagonal row publishes it, so the row swap is
+// performed by the OWNING blocks from slab data — no cross-block reads of
+// matrix memory, no fences on the bulk (G16 R1: sc1 payload -> drain ->
+// flag; consumers poll relaxed + read sc1).
+// --------------------------------------------------------------------------
+struct PanelSync2 {
+    unsigned long long key_abs[CONFLUX_PANEL_MAX_BLOCKS];
+    unsigned int key_row[CONFLUX_PANEL_MAX_BLOCKS];
+    unsigned int key_flag[CONFLUX_PANEL_MAX_BLOCKS];
+    double cand_row[CONFLUX_PANEL_MAX_BLOCKS][PANEL_NB];
+    double diag_row[PANEL_NB];
+    unsigned int diag_flag;
     unsigned int pub_flag;
+    unsigned int pub_piv;
+    unsigned int pub_win;
     unsigned int err;
 };
+
 
 typedef unsigned int __attribute__((address_space(1))) gu32;
 typedef unsigned long long __attribute__((address_space(1))) gu64;
@@ -227,143 +238,6 @@ DEVFN double ld_rlx_f64(const double *p) {
 }
 DEVFN void drain_stores() { asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); }
 
-__global__ __launch_bounds__(PANEL_TPB) void k_panel_col(
-    double *__restrict__ cm, int64_t ldc, int m, int nb, int c,
-    PanelSync *__restrict__ sync, int *__restrict__ ipiv, unsigned int epoch,
-    int nblocks) {
-    const int tid = threadIdx.x;
-    const int bid = blockIdx.x;
-    // rows of column c handled by this block: [c + bid*RPB + tid*2, ...)
-    const int base = c + bid * PANEL_RPB;
-    double val[2];
-    int row[2];
-    double amax = -1.0;
-    int arow = m;  // sentinel: larger than any row
-#pragma unroll
-    for (int q = 0; q < 2; ++q) {
-        const int r = base + tid + q * PANEL_TPB;
-        row[q] = r;
-        val[q] = (r < m) ? cm[(int64_t)c * ldc + r] : 0.0;
-        const double a = fabs(val[q]);
-        // first-max: strictly greater, or equal with smaller row index
-        if (r < m && (a > amax || (a == amax && r < arow))) { amax = a; arow = r; }
-    }
-    // block-level (abs, row) reduce with the same rule
-    __shared__ double s_abs[PANEL_TPB];
-    __shared__ int s_row[PANEL_TPB];
-    s_abs[tid] = amax;
-    s_row[tid] = arow;
-    __syncthreads();
-    for (int w = PANEL_TPB / 2; w > 0; w >>= 1) {
-        if (tid < w) {
-            const double oa = s_abs[tid + w];
-            const int orr = s_row[tid + w];
-            if (oa > s_abs[tid] || (oa == s_abs[tid] && orr < s_row[tid])) {
-                s_abs[tid] = oa;
-                s_row[tid] = orr;
-            }
-        }
-        __syncthreads();
-    }
-    if (tid == 0) {
-        union { double d; unsigned long long u; } a;
-        a.d = s_abs[0];
-        st_rlx_u64(&sync->cand_abs[bid], a.u);
-        st_rlx_u32((unsigned int *)&sync->cand_row[bid], (unsigned int)s_row[0]);
-        drain_stores();
-        st_rlx_u32(&sync->cand_flag[bid], epoch);
-    }
-
-    // ---- block 0 = reducer -------------------------------------------------
-    if (bid == 0) {
-        __shared__ int sh_piv;
-        if (tid == 0) {
-            double best = -1.0;
-            int bestrow = m;
-            unsigned spins = 0;
-            for (int b = 0; b < nblocks; ++b) {
-                while (ld_rlx_u32(&sync->cand_flag[b]) != epoch) {
-                    __builtin_amdgcn_s_sleep(1);
-                    if (++spins > 400000000u) { st_rlx_u32(&sync->err, 1u + c); goto bail;
-                    }
-                }
-                union { double d; unsigned long long u; } a;
-                a.u = ld_rlx_u64(&sync->cand_abs[b]);
-                const int rr = (int)ld_rlx_u32((unsigned int *)&sync->cand_row[b]);
-                if (a.d > best || (a.d == best && rr < bestrow)) { best = a.d; bestrow = rr; }
-            }
-            sh_piv = bestrow;
-            ipiv[0] = bestrow;   // sub-panel-absolute row of the pivot
-        bail:;
-        }
-        __syncthreads();
-        const int piv = sh_piv;
-        // swap rows c <-> piv.  Disputed cells (row piv, cols >= c) are NOT
-        // written here — the block owning row piv finalizes them from the
-        // published slab, so there is no write race.
-        //   row c   <- old row piv  (cols 0..nb)      [reducer, sc1]
-        //   row piv <- old row c    (cols 0..c-1)     [reducer, sc1]
-        //   slab rowc[j]   = old piv row value at col c+j  (new row c)
-        //   slab rowpiv[j] = old row c value at col c+j    (new row piv)
-        if (tid < nb) {
-            const int cc = tid;
-            const double oldc = cm[(int64_t)cc * ldc + c];
-            const double oldp = cm[(int64_t)cc * ldc + piv];
-            if (piv != c) {
-                st_rlx_f64(&cm[(int64_t)cc * ldc + c], oldp);
-                if (cc < c) st_rlx_f64(&cm[(int64_t)cc * ldc + piv], oldc);
-            }
-            if (cc >= c) {
-                st_rlx_f64(&sync->rowc[cc - c], oldp);
-                st_rlx_f64(&sync->rowpiv[cc - c], oldc);
-            }
-        }
-        if (tid == 0) st_rlx_u32((unsigned int *)&sync->piv_row, (unsigned int)piv);
-        drain_stores();
-        __syncthreads();
-        if (tid == 0) st_rlx_u32(&sync->pub_flag, epoch);
-    }
-
-    // ---- all blocks: wait for publication ---------------------------------
-    __shared__ double sh_rowc[PANEL_NB];   // pivot row (new row c), cols c..nb
-    __shared__ double sh_rowpiv[PANEL_NB]; // displaced row (new row piv)
-    __shared__ int sh_piv2;
-    if (tid == 0) {
-        unsigned spins = 0;
-        while (ld_rlx_u32(&sync->pub_flag) != epoch) {
-            __builtin_amdgcn_s_sleep(1);
-            if (++spins > 400000000u) { st_rlx_u32(&sync->err, 1000000u + c); break; }
-        }
-        sh_piv2 = (int)ld_rlx_u32((unsigned int *)&sync->piv_row);
-    }
-    __syncthreads();
-    if (tid < nb - c) {
-        sh_rowc[tid] = ld_rlx_f64(&sync->rowc[tid]);
-        sh_rowpiv[tid] = ld_rlx_f64(&sync->rowpiv[tid]);
-    }
-    __syncthreads();
-    const int piv = sh_piv2;
-    const double pivval = sh_rowc[0];
-    // LAPACK dgetrf2 scaling rule: multiply by reciprocal unless the pivot
-    // is exactly zero (singular: column left unscaled, as LAPACK does).
-    const double recip = (pivval != 0.0) ? 1.0 / pivval : 0.0;
-    const bool do_scale = pivval != 0.0;
-
-#pragma unroll
-    for (int q = 0; q < 2; ++q) {
-        const int r = row[q];
-        if (r >= m || r <= c) continue;   // rows strictly below the diagonal
-        double v0 = (r == piv) ? sh_rowpiv[0] : val[q];   // post-swap value
-        const double l = do_scale ? v0 * recip : v0;
-        cm[(int64_t)c * ldc + r] = l;     // final L entry (plain store)
-        for (int cc = c + 1; cc < nb; ++cc) {
-            const double prev = (r == piv) ? sh_rowpiv[cc - c]
-                                           : cm[(int64_t)cc * ldc + r];
-            cm[(int64_t)cc * ldc + r] = prev - l * sh_rowc[cc - c];
-        }
-    }
-}
-
 // ---------------------------------------------------------------------------
 // TRSM diagonal-block solvers (32-wide blocks; updates via k_dgemm_f64)
 // ---------------------------------------------------------------------------
@@ -392,6 +266,200 @@ __global__ __launch_bounds__(256) void k_trsm_left_lower_unit32(
 #pragma unroll
     for (int r = 0; r < 32; ++r)
         if (r < nb) X[r * ldx + j] = x[r];
+}
+
+// ---------------------------------------------------------------------------
+// persistent sub-panel getrf: ONE launch factors the nb(<=32)-wide sub-panel
+// with LAPACK partial pivoting (first-max rule).  Blocks own PANEL_RPB rows
+// held in LDS for all nb columns; per column the only cross-block traffic is
+// the slab handshake; the row swap is performed by the OWNING blocks from
+// slab-published rows (no cross-block matrix reads, no write races).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
+    double *__restrict__ panel, int64_t ldp, int m, int nb,
+    PanelSync2 *__restrict__ sync, int *__restrict__ ipiv,
+    unsigned int epoch0, int nblocks) {
+    const int tid = threadIdx.x, bid = blockIdx.x;
+    const int r0 = bid * PANEL_RPB + tid;
+    const int r1 = r0 + PANEL_TPB;
+    __shared__ double rows[2][PANEL_TPB][PANEL_NB + 1];
+    __shared__ double piv_lds[PANEL_NB];
+    __shared__ double diag_lds[PANEL_NB];
+    __shared__ double red_abs[PANEL_TPB];
+    __shared__ int red_row[PANEL_TPB];
+    __shared__ unsigned int sh_info[2];
+
+    for (int q = 0; q < 2; ++q) {
+        const int r = q ? r1 : r0;
+        for (int cc = 0; cc < nb; ++cc)
+            rows[q][tid][cc] = (r < m) ? panel[(int64_t)r * ldp + cc] : 0.0;
+    }
+
+    for (int c = 0; c < nb; ++c) {
+        const unsigned int epoch = epoch0 + (unsigned)c;
+        // ---- local candidate: first-max over own rows >= c --------------
+        double amax = -1.0;
+        int arow = m;
+        for (int q = 0; q < 2; ++q) {
+            const int r = q ? r1 : r0;
+            if (r >= c && r < m) {
+                const double a = fabs(rows[q][tid][c]);
+                if (a > amax || (a == amax && r < arow)) { amax = a; arow = r; }
+            }
+        }
+        // barrier-free block argmax: per-wave shuffle tree, 4 partials in
+        // LDS, every thread combines them (redundantly, no 2nd barrier)
+        for (int w = 32; w > 0; w >>= 1) {
+            const double oa = __shfl_down(amax, w);
+            const int orr = __shfl_down(arow, w);
+            if (oa > amax || (oa == amax && orr < arow)) { amax = oa; arow = orr; }
+        }
+        if ((tid & 63) == 0) {
+            red_abs[tid >> 6] = amax;
+            red_row[tid >> 6] = arow;
+        }
+        __syncthreads();
+        double wa = red_abs[0];
+        int wrow = red_row[0];
+        for (int wv = 1; wv < PANEL_TPB / 64; ++wv) {
+            const double oa = red_abs[wv];
+            const int orr = red_row[wv];
+            if (oa > wa || (oa == wa && orr < wrow)) { wa = oa; wrow = orr; }
+        }
+        // winner thread publishes its candidate row; diagonal-row owner
+        // publishes row c (both sc1 relaxed-agent, then the wave drains)
+        bool stored = false;
+        if (wrow < m && (wrow == r0 || wrow == r1)) {
+            const int q = (wrow == r1);
+            for (int cc = 0; cc < nb; ++cc)
+                st_rlx_f64(&sync->cand_row[bid][cc],
+                           q ? rows[1][tid][cc] : rows[0][tid][cc]);
+            stored = true;
+        }
+        if (c == r0 || c == r1) {
+            const int q = (c == r1);
+            for (int cc = 0; cc < nb; ++cc)
+                st_rlx_f64(&sync->diag_row[cc],
+                           q ? rows[1][tid][cc] : rows[0][tid][cc]);
+            stored = true;
+        }
+        if (stored) drain_stores();
+        __syncthreads();
+        if (tid == 0) {
+            union { double d; unsigned long long u; } a;
+            a.d = wa;
+            st_rlx_u64(&sync->key_abs[bid], a.u);
+            st_rlx_u32(&sync->key_row[bid], (unsigned)wrow);
+            drain_stores();
+            st_rlx_u32(&sync->key_flag[bid], epoch);
+            // the block owning row c raises diag_flag (its storing wave
+            // drained before the barrier above)
+            if (c >= bid * PANEL_RPB && c < (bid + 1) * PANEL_RPB)
+                st_rlx_u32(&sync->diag_flag, epoch);
+        }
+
+        // ---- reducer: block 0's first wave picks the global winner -------
+        // lane-parallel: lane b polls block b's flag, reads its candidate,
+        // then a shuffle tree applies the first-max rule across lanes
+        // (one parallel round trip instead of nblocks sequential ones).
+        if (bid == 0 && tid < 64) {
+            const int lane = tid;
+            double a_d = -1.0;
+            int a_row = m, a_win = lane;
+            for (int b = lane; b < nblocks; b += 64) {
+                unsigned spins = 0;
+                while (ld_rlx_u32(&sync->key_flag[b]) != epoch) {
+                    __builtin_amdgcn_s_sleep(1);
+                    if (++spins > 800000000u) {
+                        st_rlx_u32(&sync->err, 1u + (unsigned)c);
+                        break;
+                    }
+                }
+                union { double d; unsigned long long u; } a;
+                a.u = ld_rlx_u64(&sync->key_abs[b]);
+                const int rr = (int)ld_rlx_u32(&sync->key_row[b]);
+                if (a.d > a_d || (a.d == a_d && rr < a_row)) {
+                    a_d = a.d;
+                    a_row = rr;
+                    a_win = b;
+                }
+            }
+            for (int w = 32; w > 0; w >>= 1) {
+                const double od = __shfl_down(a_d, w);
+                const int orr = __shfl_down(a_row, w);
+                const int ow = __shfl_down(a_win, w);
+                if (od > a_d || (od == a_d && orr < a_row)) {
+                    a_d = od;
+                    a_row = orr;
+                    a_win = ow;
+                }
+            }
+            if (lane == 0) {
+                ipiv[c] = a_row;
+                st_rlx_u32(&sync->pub_piv, (unsigned)a_row);
+                st_rlx_u32(&sync->pub_win, (unsigned)a_win);
+                drain_stores();
+                st_rlx_u32(&sync->pub_flag, epoch);
+            }
+        }
+
+        // ---- all blocks: wait publication, load pivot + diag rows --------
+        if (tid == 0) {
+            unsigned spins = 0;
+            while (ld_rlx_u32(&sync->pub_flag) != epoch ||
+                   ld_rlx_u32(&sync->diag_flag) != epoch) {
+                __builtin_amdgcn_s_sleep(1);
+                if (++spins > 800000000u) {
+                    st_rlx_u32(&sync->err, 1000000u + (unsigned)c);
+                    break;
+                }
+            }
+            sh_info[0] = ld_rlx_u32(&sync->pub_piv);
+            sh_info[1] = ld_rlx_u32(&sync->pub_win);
+        }
+        __syncthreads();
+        const int piv = (int)sh_info[0];
+        const int win = (int)sh_info[1];
+        if (tid < nb) {
+            piv_lds[tid] = ld_rlx_f64(&sync->cand_row[win][tid]);
+            diag_lds[tid] = ld_rlx_f64(&sync->diag_row[tid]);
+        }
+        __syncthreads();
+        const double pivval = piv_lds[c];
+        const bool do_scale = pivval != 0.0;
+        const double recip = do_scale ? 1.0 / pivval : 0.0;
+
+        // ---- update own rows ---------------------------------------------
+        for (int q = 0; q < 2; ++q) {
+            const int r = q ? r1 : r0;
+            if (r >= m) continue;
+            double *my = rows[q][tid];
+            if (r == c) {
+                for (int cc = 0; cc < nb; ++cc) my[cc] = piv_lds[cc];
+            } else if (r == piv) {
+                for (int cc = 0; cc < c; ++cc) my[cc] = diag_lds[cc];
+                const double v0 = diag_lds[c];
+                const double l = do_scale ? v0 * recip : v0;
+                my[c] = l;
+                for (int cc = c + 1; cc < nb; ++cc)
+                    my[cc] = diag_lds[cc] - l * piv_lds[cc];
+            } else if (r > c) {
+                const double l = do_scale ? my[c] * recip : my[c];
+                my[c] = l;
+                for (int cc = c + 1; cc < nb; ++cc)
+                    my[cc] -= l * piv_lds[cc];
+            }
+        }
+        __syncthreads();
+    }
+
+    // write back (plain stores; next kernels see them at the launch boundary)
+    for (int q = 0; q < 2; ++q) {
+        const int r = q ? r1 : r0;
+        if (r >= m) continue;
+        for (int cc = 0; cc < nb; ++cc)
+            panel[(int64_t)r * ldp + cc] = rows[q][tid][cc];
+    }
 }
 
 // X (M x nb, row-major ld) <- X U^{-1} with U (nb x nb, ld ldu) upper
@@ -458,8 +526,8 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
     const int row0 = tm * GEMM_BM;
     const int64_t col0 = (int64_t)tn * GEMM_BN;
 
-    __shared__ double As[GEMM_BK][GEMM_BM + 1];   // transposed, padded
-    __shared__ double Bs[GEMM_BK][GEMM_BN + 2];
+    __shared__ double As[2][GEMM_BK][GEMM_BM + 1];   // transposed, padded
+    __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];   // double-buffered
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -501,25 +569,26 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
             rb[i] = (kk + k < K && gc < N) ? B[(int64_t)(kk + k) * ldb + gc] : 0.0;
         }
     };
-    auto write_lds = [&]() {
+    auto write_lds = [&](int buf) {
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
             const int e = tid + i * GEMM_TPB;
-            As[e & 15][e >> 4] = ra[i];
+            As[buf][e & 15][e >> 4] = ra[i];
         }
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
             const int e = tid + i * GEMM_TPB;
-            Bs[e >> 7][e & 127] = rb[i];
+            Bs[buf][e >> 7][e & 127] = rb[i];
         }
     };
 
     load_a(0);
     load_b(0);
-    write_lds();
+    write_lds(0);
+    __syncthreads();
 
+    int cur = 0;
     for (int kt = 0; kt < ktiles; ++kt) {
-        __syncthreads();                 // LDS tile kt ready
         if (kt + 1 < ktiles) {           // issue next tile's global loads
             load_a(kt + 1);
             load_b(kt + 1);
@@ -529,9 +598,9 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
             const int k = kk * 4 + fk;
             double af[4], bf[4];
 #pragma unroll
-            for (int i = 0; i < 4; ++i) af[i] = As[k][wm0 + i * 16 + frow];
+            for (int i = 0; i < 4; ++i) af[i] = As[cur][k][wm0 + i * 16 + frow];
 #pragma unroll
-            for (int j = 0; j < 4; ++j) bf[j] = Bs[k][wn0 + j * 16 + frow];
+            for (int j = 0; j < 4; ++j) bf[j] = Bs[cur][k][wn0 + j * 16 + frow];
 #pragma unroll
             for (int i = 0; i < 4; ++i)
 #pragma unroll
@@ -539,8 +608,11 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
                     acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
                         af[i], bf[j], acc[i][j], 0, 0, 0);
         }
-        __syncthreads();                 // everyone done reading tile kt
-        if (kt + 1 < ktiles) write_lds();
+        // write t+1 into the other buffer (everyone finished reading it at
+        // the barrier that ended step t-1), then one barrier per K-step
+        if (kt + 1 < ktiles) write_lds(cur ^ 1);
+        __syncthreads();
+        cur ^= 1;
     }
 
     // epilogue: C -= acc   (read-modify-write, coalesced over frag columns)
@@ -548,6 +620,132 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
     for (int i = 0; i < 4; ++i) {
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                // f64 16x16x4 C/D map (hardware-verified by mfma_probe):
+                // lane l, reg q -> D[4*q + (l>>4)][l & 15]
+                const int r = row0 + wm0 + i * 16 + q * 4 + fk;
+                const int64_t cidx = col0 + wn0 + j * 16 + frow;
+                if (r < M && cidx < N) {
+                    C[(int64_t)r * ldc + cidx] -= acc[i][j][q];
+                }
+            }
+        }
+    }
+}
+
+__global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
+    const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
+    int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
+    int ntm, int ntn) {
+    // bijective XCD swizzle (guide §5: q/r form)
+    int wg = blockIdx.x;
+    {
+        const int nwg = ntm * ntn;
+        const int q = nwg >> 3, r = nwg & 7;
+        const int xcd = wg & 7, idx = wg >> 3;
+        // inverse of dispatch round-robin: give each XCD a contiguous chunk
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+        if (nwg < 8) wg = blockIdx.x;   // tiny grids: identity
+    }
+    const int tm = wg / ntn, tn = wg % ntn;
+    const int row0 = tm * GEMM_BM;
+    const int64_t col0 = (int64_t)tn * GEMM_BN;
+
+    __shared__ double As[2][GEMM_BK][GEMM_BM + 1];   // transposed, padded
+    __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];   // double-buffered
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm0 = (wave >> 2) * 64;             // wave's 64x32 sub-tile
+    const int wn0 = (wave & 3) * 32;
+    const int frow = lane & 15;                   // fragment row/col lane part
+    const int fk = lane >> 4;                     // fragment k lane part
+
+    f64x4 acc[4][2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = f64x4{0, 0, 0, 0};
+
+    // staging registers: 8 A elements + 8 B elements per thread per K-tile
+    // (element e = tid + i*256 over the 2048-element tile; A reads coalesce
+    // over the 16-wide rows, B reads over the 128-wide rows)
+    double ra[4], rb[4];
+    const int ktiles = (K + GEMM_BK - 1) / GEMM_BK;
+
+    auto load_a = [&](int kt) {
+        const int kk = kt * GEMM_BK;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;               // 0..2047
+            const int r = e >> 4, k = e & 15;          // row-major in tile
+            const int gr = row0 + r;
+            ra[i] = (gr < M && kk + k < K) ? A[(int64_t)gr * lda + kk + k] : 0.0;
+        }
+    };
+    auto load_b = [&](int kt) {
+        const int kk = kt * GEMM_BK;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;               // 0..2047
+            const int k = e >> 7, c = e & 127;
+            const int64_t gc = col0 + c;
+            rb[i] = (kk + k < K && gc < N) ? B[(int64_t)(kk + k) * ldb + gc] : 0.0;
+        }
+    };
+    auto write_lds = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;
+            As[buf][e & 15][e >> 4] = ra[i];
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;
+            Bs[buf][e >> 7][e & 127] = rb[i];
+        }
+    };
+
+    load_a(0);
+    load_b(0);
+    write_lds(0);
+    __syncthreads();
+
+    int cur = 0;
+    for (int kt = 0; kt < ktiles; ++kt) {
+        if (kt + 1 < ktiles) {           // issue next tile's global loads
+            load_a(kt + 1);
+            load_b(kt + 1);
+        }
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            const int k = kk * 4 + fk;
+            double af[4], bf[2];
+#pragma unroll
+            for (int i = 0; i < 4; ++i) af[i] = As[cur][k][wm0 + i * 16 + frow];
+#pragma unroll
+            for (int j = 0; j < 2; ++j) bf[j] = Bs[cur][k][wn0 + j * 16 + frow];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 2; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+        // write t+1 into the other buffer (everyone finished reading it at
+        // the barrier that ended step t-1), then one barrier per K-step
+        if (kt + 1 < ktiles) write_lds(cur ^ 1);
+        __syncthreads();
+        cur ^= 1;
+    }
+
+    // epilogue: C -= acc   (read-modify-write, coalesced over frag columns)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
 #pragma unroll
             for (int q = 0; q < 4; ++q) {
                 // f64 16x16x4 C/D map (hardware-verified by mfma_probe):
@@ -675,18 +873,18 @@ void launch_cm_export(const double *cm, int64_t ldc, double *dst, int64_t ldd,
                        dim3(256), 0, s, cm, ldc, dst, ldd, rows, cols);
 }
 
-void launch_panel_col(double *cm, int64_t ldc, int m, int nb, int c,
-                      void *sync, int *ipiv, unsigned int epoch,
-                      hipStream_t s) {
-    const int rows = m - c;
-    int nblocks = (int)cdiv64(rows, PANEL_RPB);
+int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
+                        int *ipiv, unsigned int epoch0, hipStream_t s) {
+    int nblocks = (int)cdiv64(m, PANEL_RPB);
     if (nblocks < 1) nblocks = 1;
-    if (nblocks > CONFLUX_PANEL_MAX_BLOCKS) nblocks = CONFLUX_PANEL_MAX_BLOCKS;
-    hipLaunchKernelGGL(k_panel_col, dim3(nblocks), dim3(PANEL_TPB), 0, s, cm,
-                       ldc, m, nb, c, (PanelSync *)sync, ipiv, epoch, nblocks);
+    if (nblocks > CONFLUX_PANEL_MAX_BLOCKS) return -1;  // not resident: refuse
+    hipLaunchKernelGGL(k_panel_factor, dim3(nblocks), dim3(PANEL_TPB), 0, s,
+                       panel, ldp, m, nb, (PanelSync2 *)sync, ipiv, epoch0,
+                       nblocks);
+    return 0;
 }
 
-int conflux_panel_sync_bytes() { return (int)sizeof(PanelSync); }
+int conflux_panel_sync_bytes() { return (int)sizeof(PanelSync2); }
 int conflux_panel_nb() { return PANEL_NB; }
 
 void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
@@ -704,14 +902,24 @@ void launch_trsm_right_upper32(const double *U, int64_t ldu, double *X,
                        0, s, U, ldu, X, ldx, nb, M);
 }
 
+int g_dgemm_variant = -1;  // 0 = 4-wave, 1 = 8-wave; env CONFLUX_GEMM_VARIANT
+
 void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
                       int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
                       int K, hipStream_t s) {
     if (M <= 0 || N <= 0 || K <= 0) return;
+    if (g_dgemm_variant < 0) {
+        const char *e = getenv("CONFLUX_GEMM_VARIANT");
+        g_dgemm_variant = e ? atoi(e) : 1;
+    }
     const int ntm = (int)cdiv64(M, GEMM_BM);
     const int ntn = (int)cdiv64(N, GEMM_BN);
-    hipLaunchKernelGGL(k_dgemm_f64, dim3(ntm * ntn), dim3(GEMM_TPB), 0, s, A,
-                       lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
+    if (g_dgemm_variant == 1)
+        hipLaunchKernelGGL(k_dgemm_f64_w8, dim3(ntm * ntn), dim3(512), 0, s,
+                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
+    else
+        hipLaunchKernelGGL(k_dgemm_f64, dim3(ntm * ntn), dim3(GEMM_TPB), 0, s,
+                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
 }
 
 void launch_pack_candidate(const double *A10, int64_t lda, const int *gri,

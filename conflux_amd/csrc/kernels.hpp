@@ -27,8 +27,8 @@ void launch_cm_import(const double *src, int64_t lds, double *cm, int64_t ldc,
                       int rows, int cols, hipStream_t s);
 void launch_cm_export(const double *cm, int64_t ldc, double *dst, int64_t ldd,
                       int rows, int cols, hipStream_t s);
-void launch_panel_col(double *cm, int64_t ldc, int m, int nb, int c,
-                      void *sync, int *ipiv, unsigned int epoch, hipStream_t s);
+int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
+                        int *ipiv, unsigned int epoch0, hipStream_t s);
 int conflux_panel_sync_bytes();
 int conflux_panel_nb();
 void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
