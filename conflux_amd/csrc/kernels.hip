@@ -198,14 +198,10 @@ __global__ void k_cm_export(const double *__restrict__ cm, int64_t ldc,
 // --------------------------------------------------------------------------
 struct PanelSync2 {
     unsigned long long key_abs[CONFLUX_PANEL_MAX_BLOCKS];
-    unsigned int key_row[CONFLUX_PANEL_MAX_BLOCKS];
-    unsigned int key_flag[CONFLUX_PANEL_MAX_BLOCKS];
+    unsigned long long key_flag[CONFLUX_PANEL_MAX_BLOCKS];  // (epoch<<32)|row
     double cand_row[CONFLUX_PANEL_MAX_BLOCKS][PANEL_NB];
     double diag_row[PANEL_NB];
     unsigned int diag_flag;
-    unsigned int pub_flag;
-    unsigned int pub_piv;
-    unsigned int pub_win;
     unsigned int err;
 };
 
@@ -349,26 +345,27 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
             union { double d; unsigned long long u; } a;
             a.d = wa;
             st_rlx_u64(&sync->key_abs[bid], a.u);
-            st_rlx_u32(&sync->key_row[bid], (unsigned)wrow);
             drain_stores();
-            st_rlx_u32(&sync->key_flag[bid], epoch);
+            // flag granule carries the row (R2: data IS the tag)
+            st_rlx_u64(&sync->key_flag[bid],
+                       ((unsigned long long)epoch << 32) | (unsigned)wrow);
             // the block owning row c raises diag_flag (its storing wave
             // drained before the barrier above)
             if (c >= bid * PANEL_RPB && c < (bid + 1) * PANEL_RPB)
                 st_rlx_u32(&sync->diag_flag, epoch);
         }
 
-        // ---- reducer: block 0's first wave picks the global winner -------
-        // lane-parallel: lane b polls block b's flag, reads its candidate,
-        // then a shuffle tree applies the first-max rule across lanes
-        // (one parallel round trip instead of nblocks sequential ones).
-        if (bid == 0 && tid < 64) {
-            const int lane = tid;
+        // ---- EVERY block reduces the global winner itself (redundant,
+        // deterministic): lane b polls block b's flag granule, reads its
+        // |value|, shuffle-tree with the first-max rule.  Saves the central
+        // reducer's publish round trip.
+        if (tid < 64) {
             double a_d = -1.0;
-            int a_row = m, a_win = lane;
-            for (int b = lane; b < nblocks; b += 64) {
+            int a_row = m, a_win = 0;
+            for (int b = tid; b < nblocks; b += 64) {
+                unsigned long long g;
                 unsigned spins = 0;
-                while (ld_rlx_u32(&sync->key_flag[b]) != epoch) {
+                while (((g = ld_rlx_u64(&sync->key_flag[b])) >> 32) != epoch) {
                     __builtin_amdgcn_s_sleep(1);
                     if (++spins > 800000000u) {
                         st_rlx_u32(&sync->err, 1u + (unsigned)c);
@@ -377,7 +374,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                 }
                 union { double d; unsigned long long u; } a;
                 a.u = ld_rlx_u64(&sync->key_abs[b]);
-                const int rr = (int)ld_rlx_u32(&sync->key_row[b]);
+                const int rr = (int)(g & 0xffffffffu);
                 if (a.d > a_d || (a.d == a_d && rr < a_row)) {
                     a_d = a.d;
                     a_row = rr;
@@ -394,28 +391,20 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                     a_win = ow;
                 }
             }
-            if (lane == 0) {
-                ipiv[c] = a_row;
-                st_rlx_u32(&sync->pub_piv, (unsigned)a_row);
-                st_rlx_u32(&sync->pub_win, (unsigned)a_win);
-                drain_stores();
-                st_rlx_u32(&sync->pub_flag, epoch);
-            }
-        }
-
-        // ---- all blocks: wait publication, load pivot + diag rows --------
-        if (tid == 0) {
-            unsigned spins = 0;
-            while (ld_rlx_u32(&sync->pub_flag) != epoch ||
-                   ld_rlx_u32(&sync->diag_flag) != epoch) {
-                __builtin_amdgcn_s_sleep(1);
-                if (++spins > 800000000u) {
-                    st_rlx_u32(&sync->err, 1000000u + (unsigned)c);
-                    break;
+            if (tid == 0) {
+                if (bid == 0) ipiv[c] = a_row;
+                sh_info[0] = (unsigned)a_row;
+                sh_info[1] = (unsigned)a_win;
+                // wait the diagonal row publication too
+                unsigned spins = 0;
+                while (ld_rlx_u32(&sync->diag_flag) != epoch) {
+                    __builtin_amdgcn_s_sleep(1);
+                    if (++spins > 800000000u) {
+                        st_rlx_u32(&sync->err, 1000000u + (unsigned)c);
+                        break;
+                    }
                 }
             }
-            sh_info[0] = ld_rlx_u32(&sync->pub_piv);
-            sh_info[1] = ld_rlx_u32(&sync->pub_win);
         }
         __syncthreads();
         const int piv = (int)sh_info[0];
