@@ -51,15 +51,21 @@ def cpu_baseline():
     if not os.path.exists(ref):
         return None
     ncores = os.cpu_count() or 1
-    n_ranks = 4
-    omp = max(1, ncores // n_ranks)
+    # best layout from the measured sweep on this host class
+    # (tools/cpu_sweep.sh: 16 ranks x 8 OMP threads on a 4x4x1 grid beat
+    #  4x64, 64x4, 16x16 and 8x32 at N=8192)
+    n_ranks, omp, grid = 16, 8, ("4", "4", "1")
+    if ncores < n_ranks * omp:
+        n_ranks, omp = 4, max(1, ncores // 4)
+        grid = ("2", "2", "1")
     env = dict(os.environ, MKL_THREADING_LAYER="GNU",
                OMP_NUM_THREADS=str(omp), LD_LIBRARY_PATH="/opt/conda/lib")
     N, v = 8192, 512
     try:
         out = subprocess.run(
             ["/opt/conda/bin/mpiexec", "-n", str(n_ranks), ref, str(N),
-             str(v), "2", "2", "1", "-", "/tmp/confluxref_bench", "1"],
+             str(v), grid[0], grid[1], grid[2], "-", "/tmp/confluxref_bench",
+             "1"],
             env=env, capture_output=True, text=True, timeout=900)
         ms = None
         for line in out.stdout.splitlines():
@@ -71,8 +77,8 @@ def cpu_baseline():
         return {"value": round(tflops, 4), "unit": "TFLOP/s",
                 "cores": n_ranks * omp, "kind": "reference",
                 "sample": f"reference CPU path (MKL+MPICH) N={N} v={v} "
-                          f"grid 2x2x1, {n_ranks} ranks x {omp} OMP threads, "
-                          f"1 rep on this box's host cores"}
+                          f"grid {'x'.join(grid)}, {n_ranks} ranks x {omp} "
+                          f"OMP threads, 1 rep on this box's host cores"}
     except Exception:
         return None
 

@@ -234,6 +234,12 @@ DEVFN double ld_rlx_f64(const double *p) {
 }
 DEVFN void drain_stores() { asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); }
 
+typedef int i32x4 __attribute__((ext_vector_type(4)));
+union F64x2Bits {
+    double d[2];
+    i32x4 v;
+};
+
 // ---------------------------------------------------------------------------
 // TRSM diagonal-block solvers (32-wide blocks; updates via k_dgemm_f64)
 // ---------------------------------------------------------------------------
@@ -284,6 +290,11 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     __shared__ double red_abs[PANEL_TPB];
     __shared__ int red_row[PANEL_TPB];
     __shared__ unsigned int sh_info[2];
+    __shared__ int sh_pub[4];  // winner (tid, q), diag owner (tid, q)
+    const auto srsrc = __builtin_amdgcn_make_buffer_rsrc(
+        (void *)sync, (short)0, (int)sizeof(PanelSync2), 0x00020000);
+    const int cand_off = (int)offsetof(PanelSync2, cand_row) + bid * PANEL_NB * 8;
+    const int diag_off = (int)offsetof(PanelSync2, diag_row);
 
     for (int q = 0; q < 2; ++q) {
         const int r = q ? r1 : r0;
@@ -322,25 +333,40 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
             const int orr = red_row[wv];
             if (oa > wa || (oa == wa && orr < wrow)) { wa = oa; wrow = orr; }
         }
-        // winner thread publishes its candidate row; diagonal-row owner
-        // publishes row c (both sc1 relaxed-agent, then the wave drains)
-        bool stored = false;
+        // ---- cooperative publication: owners of the winner row and the
+        // diagonal row mark their (tid, q); lanes 0-15 stream the winner
+        // row, lanes 16-31 the diag row, as 16-byte sc1 buffer stores; all
+        // publishing lanes are wave 0, so one wave drain covers them.
         if (wrow < m && (wrow == r0 || wrow == r1)) {
-            const int q = (wrow == r1);
-            for (int cc = 0; cc < nb; ++cc)
-                st_rlx_f64(&sync->cand_row[bid][cc],
-                           q ? rows[1][tid][cc] : rows[0][tid][cc]);
-            stored = true;
+            sh_pub[0] = tid;
+            sh_pub[1] = (wrow == r1);
+        } else if (wrow >= m && tid == 0) {
+            sh_pub[0] = 0;  // empty block: junk row; its key_abs = -1 loses
+            sh_pub[1] = 0;
         }
         if (c == r0 || c == r1) {
-            const int q = (c == r1);
-            for (int cc = 0; cc < nb; ++cc)
-                st_rlx_f64(&sync->diag_row[cc],
-                           q ? rows[1][tid][cc] : rows[0][tid][cc]);
-            stored = true;
+            sh_pub[2] = tid;
+            sh_pub[3] = (c == r1);
         }
-        if (stored) drain_stores();
         __syncthreads();
+        if (tid < 16) {
+            F64x2Bits x;
+            x.d[0] = rows[sh_pub[1]][sh_pub[0]][2 * tid];
+            x.d[1] = rows[sh_pub[1]][sh_pub[0]][2 * tid + 1];
+            if (2 * tid < nb)
+                __builtin_amdgcn_raw_buffer_store_b128(
+                    x.v, srsrc, cand_off + 16 * tid, 0, /*sc1*/ 16);
+        } else if (tid < 32 && c >= bid * PANEL_RPB &&
+                   c < (bid + 1) * PANEL_RPB) {
+            const int l = tid - 16;
+            F64x2Bits x;
+            x.d[0] = rows[sh_pub[3]][sh_pub[2]][2 * l];
+            x.d[1] = rows[sh_pub[3]][sh_pub[2]][2 * l + 1];
+            if (2 * l < nb)
+                __builtin_amdgcn_raw_buffer_store_b128(
+                    x.v, srsrc, diag_off + 16 * l, 0, /*sc1*/ 16);
+        }
+        if (tid < 64) drain_stores();
         if (tid == 0) {
             union { double d; unsigned long long u; } a;
             a.d = wa;
@@ -409,9 +435,19 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         __syncthreads();
         const int piv = (int)sh_info[0];
         const int win = (int)sh_info[1];
-        if (tid < nb) {
-            piv_lds[tid] = ld_rlx_f64(&sync->cand_row[win][tid]);
-            diag_lds[tid] = ld_rlx_f64(&sync->diag_row[tid]);
+        if (tid < 16 && 2 * tid < nb) {
+            F64x2Bits x;
+            x.v = __builtin_amdgcn_raw_buffer_load_b128(
+                srsrc,
+                (int)offsetof(PanelSync2, cand_row) + win * PANEL_NB * 8 +
+                    16 * tid,
+                0, /*sc1*/ 16);
+            piv_lds[2 * tid] = x.d[0];
+            piv_lds[2 * tid + 1] = x.d[1];
+            x.v = __builtin_amdgcn_raw_buffer_load_b128(
+                srsrc, diag_off + 16 * tid, 0, /*sc1*/ 16);
+            diag_lds[2 * tid] = x.d[0];
+            diag_lds[2 * tid + 1] = x.d[1];
         }
         __syncthreads();
         const double pivval = piv_lds[c];
