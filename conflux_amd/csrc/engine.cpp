@@ -104,7 +104,11 @@ struct Ctx {
     bool store_factors = true;
     bool have_comm = false;
     ncclComm_t comm{};
+    ncclComm_t pcomm{};          // second comm for the lookahead panel chain
+    bool have_pcomm = false;
     hipStream_t stream{};
+    hipStream_t panel_stream{};  // lookahead stream (distributed mode)
+    hipEvent_t ev_pc{};          // panel-columns-updated event
     std::vector<RankState> rs;   // size P (sim) or 1 (distributed)
     std::vector<int> pivotInds;  // M, global pivot ids (all ranks identical)
     unsigned epoch = 1;
@@ -362,7 +366,8 @@ void plan_from_gpivots(Ctx &c, StepPlan &sp) {
     }
 }
 
-int run_step(Ctx &c, int k);
+int phase01(Ctx &c, int k, StepPlan &sp);
+int run_step(Ctx &c, int k, StepPlan &sp);
 
 int factor_loop(Ctx &c, double *elapsed_ms) {
     // reset per-factor state
@@ -397,9 +402,14 @@ int factor_loop(Ctx &c, double *elapsed_ms) {
     }
     const auto t1 = std::chrono::high_resolution_clock::now();
 
-    for (int k = 0; k < c.Nt; ++k) {
-        int rc = run_step(c, k);
+    {
+        StepPlan sp;
+        int rc = phase01(c, 0, sp);
         if (rc) return rc;
+        for (int k = 0; k < c.Nt; ++k) {
+            rc = run_step(c, k, sp);
+            if (rc) return rc;
+        }
     }
 
     HIPCHK(hipStreamSynchronize(c.stream));
@@ -427,12 +437,11 @@ int factor_loop(Ctx &c, double *elapsed_ms) {
     return 0;
 }
 
-int run_step(Ctx &c, int k) {
+int phase01(Ctx &c, int k, StepPlan &sp) {
     const int v = c.v, Px = c.Px, Py = c.Py, Pz = c.Pz;
     const int64_t Nl = c.Nl;
     const int loff = (k / Py) * v;
     const int kcol = k % Py, krow = k % Px;
-    const int64_t wA01 = Nl - loff;
     const int n_rounds = Px > 1 ? (int)std::ceil(std::log2((double)Px)) : 0;
 
     // ---- step 0: copy active col block to A10, depth-reduce (C1) ----------
@@ -463,7 +472,6 @@ int run_step(Ctx &c, int k) {
     }
 
     // ---- step 1: tournament pivoting --------------------------------------
-    StepPlan sp;
     sp.gpivots.assign(v, -1);
     std::vector<int> perm;
     for (auto &r : c.rs) {
@@ -511,9 +519,9 @@ int run_step(Ctx &c, int k) {
                 const int64_t roff = half - soff;
                 NCCLCHK(ncclGroupStart());
                 NCCLCHK(ncclSend(me.cand + soff, half, ncclDouble,
-                                 grank_of(c, src, kcol, 0), c.comm, c.stream));
+                                 grank_of(c, src, kcol, 0), c.pcomm, c.stream));
                 NCCLCHK(ncclRecv(me.cand + roff, half, ncclDouble,
-                                 grank_of(c, src, kcol, 0), c.comm, c.stream));
+                                 grank_of(c, src, kcol, 0), c.pcomm, c.stream));
                 NCCLCHK(ncclGroupEnd());
             }
         } else {
@@ -568,13 +576,13 @@ int run_step(Ctx &c, int k) {
             if (me.pj == kcol && me.pk == 0 &&
                 !(me.pi == krow && me.pj == kcol)) {
                 NCCLCHK(ncclSend(me.A00, i64(v) * v, ncclDouble,
-                                 grank_of(c, krow, me.pi, 0), c.comm,
+                                 grank_of(c, krow, me.pi, 0), c.pcomm,
                                  c.stream));
             }
             if (me.pi == krow && me.pk == 0 &&
                 !(me.pj == kcol)) {
                 NCCLCHK(ncclRecv(me.A00, i64(v) * v, ncclDouble,
-                                 grank_of(c, me.pj, kcol, 0), c.comm,
+                                 grank_of(c, me.pj, kcol, 0), c.pcomm,
                                  c.stream));
             }
             // gpivots: root (pi, kcol, 0) -> its jk plane
@@ -583,12 +591,12 @@ int run_step(Ctx &c, int k) {
                     for (int pk = 0; pk < Pz; ++pk) {
                         if (pj == kcol && pk == 0) continue;
                         NCCLCHK(ncclSend(me.d_gpivots, v, ncclInt32,
-                                         grank_of(c, me.pi, pj, pk), c.comm,
+                                         grank_of(c, me.pi, pj, pk), c.pcomm,
                                          c.stream));
                     }
             } else {
                 NCCLCHK(ncclRecv(me.d_gpivots, v, ncclInt32,
-                                 grank_of(c, me.pi, kcol, 0), c.comm,
+                                 grank_of(c, me.pi, kcol, 0), c.pcomm,
                                  c.stream));
             }
             NCCLCHK(ncclGroupEnd());
@@ -613,6 +621,22 @@ int run_step(Ctx &c, int k) {
 
     plan_from_gpivots(c, sp);
     std::copy_n(sp.gpivots.begin(), v, c.pivotInds.begin() + i64(k) * v);
+    return 0;
+}
+
+// steps 2..6 of superstep k; sp holds this step's pivot plan (from phase01).
+// At the tail, the trailing update is split so that the columns of step
+// k+1's panel are updated FIRST, and phase01(k+1) runs on the panel stream
+// CONCURRENTLY with the rest of the trailing update (lookahead — numerically
+// identical: the column split does not reorder any K-sum).  On exit sp holds
+// step k+1's plan.
+int run_step(Ctx &c, int k, StepPlan &sp) {
+    const int v = c.v, Px = c.Px, Py = c.Py, Pz = c.Pz;
+    const int64_t Nl = c.Nl;
+    const int loff = (k / Py) * v;
+    const int kcol = k % Py, krow = k % Px;
+    const int64_t wA01 = Nl - loff;
+    (void)kcol;
 
     // ---- step 2: push pivot rows up, pack, depth-reduce (C7) --------------
     for (auto &r : c.rs) {
@@ -958,16 +982,49 @@ int run_step(Ctx &c, int k) {
         }
     }
 
-    // ---- step 6: trailing update (the flop carrier) ------------------------
-    for (auto &r : c.rs) {
-        if (r.nact <= 0) continue;
-        const double fl = 2.0 * r.nact * (double)wA01 * c.nlayr;
+    // ---- step 6: trailing update (the flop carrier), split for lookahead --
+    auto gemm_piece = [&](RankState &r, int64_t col0, int64_t ncols) -> int {
+        if (r.nact <= 0 || ncols <= 0) return 0;
+        const double fl = 2.0 * r.nact * (double)ncols * c.nlayr;
         size_t slot;
         if (ev_begin(c, 0, fl, &slot)) return CONFLUX_LU_EHIP;
-        launch_dgemm_f64(r.A10Rcv, c.nlayr, r.A01Rcv, Nl,
-                         r.A11 + i64(r.fnp) * Nl + loff, Nl, r.nact, wA01,
+        launch_dgemm_f64(r.A10Rcv, c.nlayr, r.A01Rcv + (col0 - loff), Nl,
+                         r.A11 + i64(r.fnp) * Nl + col0, Nl, r.nact, ncols,
                          c.nlayr, c.stream);
-        if (ev_end(c, slot)) return CONFLUX_LU_EHIP;
+        return ev_end(c, slot) ? CONFLUX_LU_EHIP : 0;
+    };
+    const bool look = (k + 1 < c.Nt);
+    const int ncol = (k + 1) % Py;
+    const int64_t lnext = i64(v) * ((k + 1) / Py);
+    // (a) the columns step k+1's panel needs, first
+    if (look)
+        for (auto &r : c.rs)
+            if (r.pj == ncol)
+                if (gemm_piece(r, lnext, v)) return CONFLUX_LU_EHIP;
+    if (look && !c.sim && c.panel_stream) {
+        HIPCHK(hipEventRecord(c.ev_pc, c.stream));
+    }
+    // (b) the rest of the trailing update
+    for (auto &r : c.rs) {
+        if (look && r.pj == ncol) {
+            if (gemm_piece(r, loff, lnext - loff)) return CONFLUX_LU_EHIP;
+            if (gemm_piece(r, lnext + v, Nl - (lnext + v))) return CONFLUX_LU_EHIP;
+        } else {
+            if (gemm_piece(r, loff, wA01)) return CONFLUX_LU_EHIP;
+        }
+    }
+    // (c) lookahead: step k+1's panel chain, overlapped with (b)
+    if (look) {
+        hipStream_t saved = c.stream;
+        if (!c.sim && c.panel_stream) {
+            HIPCHK(hipStreamWaitEvent(c.panel_stream, c.ev_pc, 0));
+            c.stream = c.panel_stream;
+        }
+        StepPlan nsp;
+        int rc = phase01(c, k + 1, nsp);
+        c.stream = saved;
+        if (rc) return rc;
+        sp = std::move(nsp);
     }
     return 0;
 }
@@ -1037,6 +1094,11 @@ int conflux_lu_create(int N, int v, int Px, int Py, int Pz, int rank,
         c->rs.resize(1);
         const int pi = rank / (Py * Pz), pj = (rank / Pz) % Py, pk = rank % Pz;
         if (alloc_rank(*c, c->rs[0], pi, pj, pk)) { delete c; return CONFLUX_LU_EHIP; }
+        if (hipStreamCreate(&c->panel_stream) != hipSuccess ||
+            hipEventCreate(&c->ev_pc) != hipSuccess) {
+            delete c;
+            return CONFLUX_LU_EHIP;
+        }
         if (world > 1) {
             if (!nccl_uid) { delete c; return CONFLUX_LU_EARG; }
             ncclUniqueId id;
@@ -1046,6 +1108,12 @@ int conflux_lu_create(int N, int v, int Px, int Py, int Pz, int rank,
                 return CONFLUX_LU_ECOMM;
             }
             c->have_comm = true;
+            if (ncclCommSplit(c->comm, 0, rank, &c->pcomm, nullptr) !=
+                ncclSuccess) {
+                delete c;
+                return CONFLUX_LU_ECOMM;
+            }
+            c->have_pcomm = true;
         }
     }
     *out = c;
@@ -1259,7 +1327,10 @@ int conflux_lu_destroy(conflux_lu_ctx *c) {
         (void)hipEventDestroy(e.a);
         (void)hipEventDestroy(e.b);
     }
+    if (c->have_pcomm) (void)ncclCommDestroy(c->pcomm);
     if (c->have_comm) (void)ncclCommDestroy(c->comm);
+    if (c->panel_stream) (void)hipStreamDestroy(c->panel_stream);
+    if (c->ev_pc) (void)hipEventDestroy(c->ev_pc);
     (void)hipStreamDestroy(c->stream);
     delete c;
     return CONFLUX_LU_OK;
