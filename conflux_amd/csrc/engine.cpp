@@ -22,6 +22,7 @@
 #include <chrono>
 #include <cmath>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <functional>
 #include <numeric>
@@ -73,6 +74,7 @@ struct RankState {
     double *Fres = nullptr;     // Ml x Nl (store_factors only)
     double *A10hist = nullptr;  // Ml x Nl (store_factors only)
     int *d_ipiv = nullptr;      // v
+    int *d_swap = nullptr;      // 128: laswp dst/src row maps
     int *d_idx = nullptr;       // 4v: pivot rows / early / late / order
     int *d_gri = nullptr;       // Ml
     int *d_gpivots = nullptr;   // v
@@ -144,6 +146,7 @@ int alloc_rank(Ctx &c, RankState &r, int pi, int pj, int pk) {
     HIPCHK(hipMalloc(&r.redtmp, i64(std::max(1, c.Pz - 1)) * Ml * v * 8));
     HIPCHK(hipMalloc(&r.slabs, Ml * v * 8));
     HIPCHK(hipMalloc(&r.d_ipiv, (v + 8) * 4));
+    HIPCHK(hipMalloc(&r.d_swap, 128 * 4));
     HIPCHK(hipMalloc(&r.d_idx, 4 * v * 4));
     HIPCHK(hipMalloc(&r.d_gri, Ml * 4));
     HIPCHK(hipMalloc(&r.d_gpivots, v * 4));
@@ -167,7 +170,7 @@ void free_rank(RankState &r) {
                       r.panel, r.cm, r.A01pack, r.rowtmp, r.redtmp, r.slabs,
                       r.Fres, r.A10hist})
         if (p) (void)hipFree(p);
-    for (int *p : {r.d_ipiv, r.d_idx, r.d_gri, r.d_gpivots, r.d_perm})
+    for (int *p : {r.d_ipiv, r.d_swap, r.d_idx, r.d_gri, r.d_gpivots, r.d_perm})
         if (p) (void)hipFree(p);
     if (r.sync) (void)hipFree(r.sync);
 }
@@ -215,9 +218,15 @@ int factor_panel(Ctx &c, RankState &r, int n, std::vector<int> &ipiv_out) {
             return CONFLUX_LU_EINTERNAL;
         }
         c.epoch += nb;
-        // apply the sub-panel's swaps to the rest of the panel width
-        launch_laswp(r.panel, v, jb, r.d_ipiv + jb, nb, 0, jb, c.stream);
-        launch_laswp(r.panel, v, jb, r.d_ipiv + jb, nb, jb + nb, v, c.stream);
+        // apply the sub-panel's swaps to the rest of the panel width:
+        // build the realized row permutation once, then two parallel passes
+        // over the non-sub-panel columns (vs nswap serialized round trips)
+        if (v > nb) {
+            launch_swap_map(r.d_ipiv + jb, nb, jb, r.d_swap, r.d_swap + 64,
+                            c.stream);
+            launch_rowperm_skip(r.panel, v, r.d_swap, r.d_swap + 64, 2 * nb,
+                                jb, nb, v - nb, r.rowtmp, c.stream);
+        }
         if (jb + nb < v && m > nb) {
             // U block: rows jb..jb+nb of cols jb+nb..v
             launch_trsm_left_lower_unit32(r.panel + i64(jb) * v + jb, v,
@@ -1013,10 +1022,17 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
             if (gemm_piece(r, loff, wA01)) return CONFLUX_LU_EHIP;
         }
     }
-    // (c) lookahead: step k+1's panel chain, overlapped with (b)
+    // (c) lookahead: step k+1's panel chain, overlapped with (b).
+    // The overlap engages only for world > 1 (it frees non-participant
+    // ranks); on a single GPU the latency-bound panel handshake slows more
+    // under GEMM load than the overlap saves (measured: 256 vs 251 ms at
+    // N=16384), so world == 1 keeps the sequential order on one stream.
     if (look) {
         hipStream_t saved = c.stream;
-        if (!c.sim && c.panel_stream) {
+        const char *lk = getenv("CONFLUX_LOOKAHEAD");
+        const bool async_look =
+            !c.sim && c.panel_stream && (lk ? atoi(lk) != 0 : c.world > 1);
+        if (async_look) {
             HIPCHK(hipStreamWaitEvent(c.panel_stream, c.ev_pc, 0));
             c.stream = c.panel_stream;
         }
@@ -1280,6 +1296,9 @@ int conflux_lu_debug_getrf(int n, int v, double *panel, int *ipiv_out) {
     RankState r;
     HIPCHK(hipMalloc(&r.panel, i64(std::max(n, 2 * v)) * v * 8));
     HIPCHK(hipMalloc(&r.d_ipiv, (v + 8) * 4));
+    HIPCHK(hipMalloc(&r.d_swap, 128 * 4));
+    HIPCHK(hipMalloc(&r.rowtmp, i64(64) * v * 8));
+    HIPCHK(hipMalloc(&r.d_swap, 128 * 4));
     HIPCHK(hipMalloc(&r.sync, conflux_panel_sync_bytes()));
     HIPCHK(hipMemset(r.sync, 0, conflux_panel_sync_bytes()));
     HIPCHK(hipMemcpy(r.panel, panel, i64(n) * v * 8, hipMemcpyHostToDevice));
@@ -1291,6 +1310,7 @@ int conflux_lu_debug_getrf(int n, int v, double *panel, int *ipiv_out) {
         std::copy(ipiv.begin(), ipiv.end(), ipiv_out);
     }
     (void)hipFree(r.panel); (void)hipFree(r.d_ipiv);
+    (void)hipFree(r.d_swap); (void)hipFree(r.rowtmp);
     (void)hipFree(r.sync);
     for (auto &e : c.evs) { (void)hipEventDestroy(e.a); (void)hipEventDestroy(e.b); }
     (void)hipStreamDestroy(c.stream);

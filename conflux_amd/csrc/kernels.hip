@@ -135,6 +135,87 @@ __global__ void k_laswp(double *__restrict__ A, int64_t lda, int i0,
     }
 }
 
+// Build the row-permutation the dlaswp swap sequence realizes: after
+// applying swaps (i0+s <-> i0+piv[s]) in order, row dst_idx[r] holds the
+// original content of row src_idx[r].  2*nb entries, identity-padded.
+// Single-thread kernel (<= 64*64 scalar ops) so laswp becomes two parallel
+// passes instead of nswap serialized global round trips per column.
+__global__ void k_swap_map(const int *__restrict__ piv, int nb, int i0,
+                           int *__restrict__ dst_idx,
+                           int *__restrict__ src_idx) {
+    // LDS state (a private array would spill to scratch — rule 20):
+    // rows [i0, i0+nb) are direct-indexed; rows below the sub-panel head go
+    // to a small overflow list (at most nb entries).
+    __shared__ int base_src[32];
+    __shared__ int ovf_pos[32], ovf_src[32];
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    for (int i = 0; i < nb; ++i) base_src[i] = i0 + i;
+    int novf = 0;
+    for (int s = 0; s < nb; ++s) {
+        const int p = i0 + piv[s];
+        int *b;
+        if (p < i0 + nb) {
+            b = &base_src[p - i0];
+        } else {
+            int j = 0;
+            for (; j < novf && ovf_pos[j] != p; ++j) {
+            }
+            if (j == novf) {
+                ovf_pos[j] = p;
+                ovf_src[j] = p;
+                ++novf;
+            }
+            b = &ovf_src[j];
+        }
+        const int t = base_src[s];
+        base_src[s] = *b;
+        *b = t;
+    }
+    for (int i = 0; i < 2 * nb; ++i) {
+        // pad by duplicating entry 0 (same dst, same src: benign identical
+        // concurrent writes) so the launch geometry can be static
+        int d, sv;
+        if (i < nb) {
+            d = i0 + i;
+            sv = base_src[i];
+        } else if (i - nb < novf) {
+            d = ovf_pos[i - nb];
+            sv = ovf_src[i - nb];
+        } else {
+            d = i0;
+            sv = base_src[0];
+        }
+        dst_idx[i] = d;
+        src_idx[i] = sv;
+    }
+}
+
+// gather/scatter with a skipped column range [skip0, skip0+skipn): the
+// sub-panel columns were already swapped inside k_panel_factor.
+__global__ void k_rowperm_gather_skip(const double *__restrict__ src,
+                                      int64_t lds, double *__restrict__ tmp,
+                                      const int *__restrict__ src_idx,
+                                      int n_rows, int64_t skip0, int64_t skipn,
+                                      int64_t tot_cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)n_rows * tot_cols) return;
+    const int64_t r = i / tot_cols, cc = i % tot_cols;
+    const int64_t c = (cc < skip0) ? cc : cc + skipn;
+    tmp[r * tot_cols + cc] = src[(int64_t)src_idx[r] * lds + c];
+}
+
+__global__ void k_rowperm_scatter_skip(const double *__restrict__ tmp,
+                                       double *__restrict__ dst, int64_t ldd,
+                                       const int *__restrict__ dst_idx,
+                                       int n_rows, int64_t skip0,
+                                       int64_t skipn, int64_t tot_cols) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (int64_t)n_rows * tot_cols) return;
+    const int64_t r = i / tot_cols, cc = i % tot_cols;
+    const int64_t c = (cc < skip0) ? cc : cc + skipn;
+    dst[(int64_t)dst_idx[r] * ldd + c] = tmp[r * tot_cols + cc];
+}
+
 // dst row dst_idx[i] <- src row src_idx[i]; row sets must be disjoint
 // (push_pivots_up phase 2: early non-pivots into vacated late-pivot slots)
 __global__ void k_row_move(const double *__restrict__ src, int64_t lds,
@@ -896,6 +977,26 @@ void launch_cm_export(const double *cm, int64_t ldc, double *dst, int64_t ldd,
     if (rows <= 0 || cols <= 0) return;
     hipLaunchKernelGGL(k_cm_export, dim3(cdiv64((int64_t)rows * cols, 256)),
                        dim3(256), 0, s, cm, ldc, dst, ldd, rows, cols);
+}
+
+void launch_swap_map(const int *piv, int nb, int i0, int *dst_idx,
+                     int *src_idx, hipStream_t s) {
+    hipLaunchKernelGGL(k_swap_map, dim3(1), dim3(64), 0, s, piv, nb, i0,
+                       dst_idx, src_idx);
+}
+
+void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
+                         const int *src_idx, int n_rows, int64_t skip0,
+                         int64_t skipn, int64_t tot_cols, double *tmp,
+                         hipStream_t s) {
+    if (n_rows <= 0 || tot_cols <= 0) return;
+    const int64_t n = (int64_t)n_rows * tot_cols;
+    hipLaunchKernelGGL(k_rowperm_gather_skip, dim3(cdiv64(n, 256)), dim3(256),
+                       0, s, mat, ld, tmp, src_idx, n_rows, skip0, skipn,
+                       tot_cols);
+    hipLaunchKernelGGL(k_rowperm_scatter_skip, dim3(cdiv64(n, 256)), dim3(256),
+                       0, s, tmp, mat, ld, dst_idx, n_rows, skip0, skipn,
+                       tot_cols);
 }
 
 int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,

@@ -27,6 +27,12 @@ void launch_cm_import(const double *src, int64_t lds, double *cm, int64_t ldc,
                       int rows, int cols, hipStream_t s);
 void launch_cm_export(const double *cm, int64_t ldc, double *dst, int64_t ldd,
                       int rows, int cols, hipStream_t s);
+void launch_swap_map(const int *piv, int nb, int i0, int *dst_idx,
+                     int *src_idx, hipStream_t s);
+void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
+                         const int *src_idx, int n_rows, int64_t skip0,
+                         int64_t skipn, int64_t tot_cols, double *tmp,
+                         hipStream_t s);
 int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
                         int *ipiv, unsigned int epoch0, hipStream_t s);
 int conflux_panel_sync_bytes();
