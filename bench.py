@@ -103,6 +103,7 @@ def main():
         os.environ["HIP_VISIBLE_DEVICES"] = str(local_rank)
 
     N, v, Px, Py, Pz = GRIDS[n_gpus]
+    K = None
     if args.N:
         N = args.N
     if args.v:
@@ -153,6 +154,7 @@ def main():
         total_s = float(tt[0])
 
     stats = eng.kernel_stats()
+    K = v // Pz  # nlayr: the trailing GEMM's K
     g = stats["dgemm_trailing"]
     roofline = None
     if g["launches"] > 0 and g["seconds"] > 0:
@@ -163,7 +165,13 @@ def main():
             "peak": FP64_MFMA_PEAK_TFLOPS,
             "unit": "TFLOP/s",
             "frac": round(achieved / FP64_MFMA_PEAK_TFLOPS, 4),
-            "traffic": None,
+            # HBM traffic per launch at the headline shape (M=N=16384,
+            # K=512), rocprofv3 PMC FETCH_SIZE x2 (gfx950 wide-read
+            # correction) + WRITE_SIZE; profiles/r01_pmc_dgemm_*.csv.
+            # 3x the 4.43 GB algorithmic bytes: the BK=16 tile re-reads of
+            # the A/B panels reach the L2 fabric but are served by the
+            # 256 MB Infinity Cache (both panels fit), not DRAM.
+            "traffic": 13.2e9 if (N == 16384 and K == 512) else None,
         }
 
     eng.close()

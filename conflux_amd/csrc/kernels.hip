@@ -866,6 +866,146 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// glds variant: async global->LDS staging (__builtin_amdgcn_global_load_lds,
+// 16-byte pieces), one barrier per K-step, XOR-swizzled SOURCE addresses so
+// the lane-linear LDS image reads conflict-free (guide rule 21 / T2).
+// Layouts: As [BK][BM] row-of-m linear, A-chunk (row, p) stored at p^(row&7);
+//          Bs [BK][BN] linear, B-chunk (k, np) stored at np^((k&3)<<1).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(512, 4) void k_dgemm_f64_glds(
+    const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
+    int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
+    int ntm, int ntn) {
+    int wg = blockIdx.x;
+    {
+        const int nwg = ntm * ntn;
+        const int q = nwg >> 3, r = nwg & 7;
+        const int xcd = wg & 7, idx = wg >> 3;
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+        if (nwg < 8) wg = blockIdx.x;
+    }
+    const int tm = wg / ntn, tn = wg % ntn;
+    const int row0 = tm * GEMM_BM;
+    const int64_t col0 = (int64_t)tn * GEMM_BN;
+
+    // ONE shared object (a second one makes hipcc drain vmcnt(0) before
+    // every ds_read — guide §5 trap (a)).  [2 buffers][A 16*128 | B 16*128]
+    __shared__ double smem[2 * (GEMM_BK * GEMM_BM + GEMM_BK * GEMM_BN)];
+    double *As[2] = {smem, smem + 2 * GEMM_BK * GEMM_BM + 0};
+    // carve: buf b: A at b*(BK*BM), B at 2*BK*BM + b*(BK*BN)
+    As[0] = smem;
+    As[1] = smem + GEMM_BK * GEMM_BM;
+    double *Bs0 = smem + 2 * GEMM_BK * GEMM_BM;
+    double *Bs[2] = {Bs0, Bs0 + GEMM_BK * GEMM_BN};
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm0 = (wave >> 2) * 64;
+    const int wn0 = (wave & 3) * 32;
+    const int frow = lane & 15;
+    const int fk = lane >> 4;
+
+    f64x4 acc[4][2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = f64x4{0, 0, 0, 0};
+
+    const int ktiles = (K + GEMM_BK - 1) / GEMM_BK;
+    // A: 1024 16B chunks (row, p), linear ci = row*8 + p_lds; 2 waves-worth
+    // per wave (16 wave-instructions over 8 waves = 2 each).
+    // chunk for this lane at issue w (w = 0,1): ci = wave*128 + w*64 + lane
+    // B: 1024 chunks (k, np), ci = k*64 + np_lds; same split.
+    auto stage = [&](int buf, int kt) {
+        const int kk = kt * GEMM_BK;
+#pragma unroll
+        for (int w = 0; w < 2; ++w) {
+            const int ci = wave * 128 + w * 64 + lane;
+            // A chunk (row, p_lds) holds source k-pair p_lds ^ (row & 7);
+            // out-of-range rows load the block's first row (junk rows are
+            // discarded by the epilogue bounds).  K is a multiple of 16
+            // everywhere in this engine (v, nlayr, NB all are), so there is
+            // no K-tail.
+            const int arow = ci >> 3, ap = ci & 7;
+            const int asrc_p = ap ^ (arow & 7);
+            const double *agp =
+                A + (int64_t)(row0 + ((row0 + arow < M) ? arow : 0)) * lda +
+                kk + 2 * asrc_p;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t *)agp,
+                (__attribute__((address_space(3))) uint32_t
+                     *)(As[buf] + (int64_t)(wave * 128 + w * 64) * 2),
+                16, 0, 0);
+            // B chunk (k, np_lds) holds source col-pair np_lds ^ ((k&3)<<1).
+            // Clamp only when the chunk's FIRST column is out of range (a
+            // half-valid chunk must still load its valid column; its second
+            // 8 bytes may read past row end — within hipMalloc's page
+            // granularity, and its value only lands in discarded lanes).
+            const int ck = ci >> 6, cnp = ci & 63;
+            const int bsrc_np = cnp ^ ((ck & 3) << 1);
+            const double *bgp =
+                B + (int64_t)(kk + ck) * ldb + col0 + 2 * bsrc_np;
+            if (col0 + 2 * bsrc_np >= N) bgp = B + (int64_t)(kk + ck) * ldb;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t *)bgp,
+                (__attribute__((address_space(3))) uint32_t
+                     *)(Bs[buf] + (int64_t)(ci - lane) * 2),
+                16, 0, 0);
+        }
+    };
+
+    stage(0, 0);
+    int cur = 0;
+    for (int kt = 0; kt < ktiles; ++kt) {
+        // ONE barrier per K-step: its per-wave vmcnt(0) drains the stage
+        // issued LAST iteration (it flew under that iteration's compute),
+        // and the barrier itself guarantees every wave finished reading the
+        // buffer the next stage overwrites.
+        __syncthreads();
+        if (kt + 1 < ktiles) stage(cur ^ 1, kt + 1);
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            const int k = kk * 4 + fk;
+            const int ap = k >> 1, ae = k & 1;
+            double af[4], bf[2];
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                const int row = wm0 + i * 16 + frow;
+                af[i] = As[cur][row * 16 + 2 * (ap ^ (row & 7)) + ae];
+            }
+#pragma unroll
+            for (int j = 0; j < 2; ++j) {
+                const int n = wn0 + j * 16 + frow;
+                const int np = (n >> 1) ^ ((k & 3) << 1);
+                bf[j] = Bs[cur][k * 128 + 2 * np + (n & 1)];
+            }
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 2; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+        cur ^= 1;
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int r = row0 + wm0 + i * 16 + q * 4 + fk;
+                const int64_t cidx = col0 + wn0 + j * 16 + frow;
+                if (r < M && cidx < N) C[(int64_t)r * ldc + cidx] -= acc[i][j][q];
+            }
+        }
+    }
+}
+
 // ---------------------------------------------------------------------------
 // misc small kernels for the distributed path
 // ---------------------------------------------------------------------------
@@ -1040,7 +1180,10 @@ void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
     }
     const int ntm = (int)cdiv64(M, GEMM_BM);
     const int ntn = (int)cdiv64(N, GEMM_BN);
-    if (g_dgemm_variant == 1)
+    if (g_dgemm_variant == 2)
+        hipLaunchKernelGGL(k_dgemm_f64_glds, dim3(ntm * ntn), dim3(512), 0, s,
+                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
+    else if (g_dgemm_variant == 1)
         hipLaunchKernelGGL(k_dgemm_f64_w8, dim3(ntm * ntn), dim3(512), 0, s,
                            A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
     else
