@@ -107,6 +107,7 @@ struct Ctx {
     bool have_comm = false;
     ncclComm_t comm{};
     ncclComm_t pcomm{};          // second comm for the lookahead panel chain
+    ncclComm_t acomm{};          // the comm the CURRENT phase must use
     bool have_pcomm = false;
     hipStream_t stream{};
     hipStream_t panel_stream{};  // lookahead stream (distributed mode)
@@ -325,12 +326,12 @@ int reduce_over_pk(Ctx &c, int pi, int pj, int64_t count,
     if (me.pi != pi || me.pj != pj) return 0;
     if (me.pk != 0) {
         NCCLCHK(ncclSend(bufof(me), count, ncclDouble,
-                         grank_of(c, pi, pj, 0), c.comm, c.stream));
+                         grank_of(c, pi, pj, 0), c.acomm, c.stream));
     } else {
         for (int pk = 1; pk < c.Pz; ++pk) {
             double *tmp = me.redtmp + i64(pk - 1) * c.Ml * c.v;
             NCCLCHK(ncclRecv(tmp, count, ncclDouble, grank_of(c, pi, pj, pk),
-                             c.comm, c.stream));
+                             c.acomm, c.stream));
         }
         // adds must come after the group closes; caller handles via
         // reduce_over_pk_finish
@@ -447,6 +448,7 @@ int factor_loop(Ctx &c, double *elapsed_ms) {
 }
 
 int phase01(Ctx &c, int k, StepPlan &sp) {
+    c.acomm = c.pcomm;  // the panel chain's comm (may overlap main-stream ops)
     const int v = c.v, Px = c.Px, Py = c.Py, Pz = c.Pz;
     const int64_t Nl = c.Nl;
     const int loff = (k / Py) * v;
@@ -640,6 +642,7 @@ int phase01(Ctx &c, int k, StepPlan &sp) {
 // identical: the column split does not reorder any K-sum).  On exit sp holds
 // step k+1's plan.
 int run_step(Ctx &c, int k, StepPlan &sp) {
+    c.acomm = c.comm;
     const int v = c.v, Px = c.Px, Py = c.Py, Pz = c.Pz;
     const int64_t Nl = c.Nl;
     const int loff = (k / Py) * v;

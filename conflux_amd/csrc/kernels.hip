@@ -4,15 +4,16 @@
 //   k_dgemm_f64        <- trailing-update cblas_dgemm (conflux_opt.hpp:1628-1633)
 //                         and the panel-update GEMMs of blocked getrf/TRSM;
 //                         hand-written v_mfma_f64_16x16x4_f64, LDS-staged.
-//   k_panel_col        <- one column of LAPACKE_dgetrf partial pivoting
-//                         (conflux_opt.hpp:143-166 LUP): grid-wide first-max
-//                         argmax + row swap + scale + rank-1 on a col-major
-//                         sub-panel; agent-scope release/acquire handshake.
+//   k_panel_factor     <- LAPACKE_dgetrf partial pivoting
+//                         (conflux_opt.hpp:143-166 LUP): persistent sub-panel
+//                         kernel, LDS-resident rows, grid-wide first-max
+//                         argmax per column via an agent-scope
+//                         release/acquire slab handshake.
 //   k_trsm_*           <- cblas_dtrsm Right/Upper/NonUnit (:1347) and
 //                         Left/Lower/Unit (:1539), 32-wide diagonal blocks +
 //                         k_dgemm_f64 updates.
-//   k_row_gather/scatter, k_laswp, k_copy2d, ...
-//                      <- push_pivots_up / permute_rows / mcopy
+//   k_row_gather/scatter/move, k_swap_map/k_rowperm_*, k_copy2d, ...
+//                      <- push_pivots_up / permute_rows / dlaswp / mcopy
 //                         (conflux_opt.hpp:176-218, utils.hpp:48-160,
 //                          memory_utils.hpp:8-34) as coalesced index-vector
 //                         kernels.
@@ -117,23 +118,6 @@ __global__ void k_row_scatter(const double *__restrict__ src, int64_t lds,
 // src rows idx[i] copied to dst rows dst_idx[i] within SAME buffer is unsafe;
 // engine always stages through separate buffers (3-phase like the reference).
 
-// Apply nswap sequential row swaps (i0+s <-> i0+piv[s]) to columns
-// [c0, c1) of a row-major matrix — the dlaswp step of blocked getrf.
-// piv entries are relative to row i0 (what k_panel_col records).
-// Each thread owns one column and performs the swap sequence in order.
-__global__ void k_laswp(double *__restrict__ A, int64_t lda, int i0,
-                        const int *__restrict__ piv, int nswap, int64_t c0,
-                        int64_t c1) {
-    const int64_t c = c0 + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (c >= c1) return;
-    for (int s = 0; s < nswap; ++s) {
-        const int64_t r1 = i0 + s, r2 = i0 + piv[s];
-        if (r1 == r2) continue;
-        const double t = A[r1 * lda + c];
-        A[r1 * lda + c] = A[r2 * lda + c];
-        A[r2 * lda + c] = t;
-    }
-}
 
 // Build the row-permutation the dlaswp swap sequence realizes: after
 // applying swaps (i0+s <-> i0+piv[s]) in order, row dst_idx[r] holds the
@@ -229,24 +213,7 @@ __global__ void k_row_move(const double *__restrict__ src, int64_t lds,
     dst[(int64_t)dst_idx[r] * ldd + c] = src[(int64_t)src_idx[r] * lds + c];
 }
 
-// row-major (rows x cols, ld) -> col-major scratch (ldc >= rows)
-__global__ void k_cm_import(const double *__restrict__ src, int64_t lds,
-                            double *__restrict__ cm, int64_t ldc, int rows,
-                            int cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)rows * cols) return;
-    const int64_t r = i / cols, c = i % cols;   // read coalesced over c
-    cm[c * ldc + r] = src[r * lds + c];
-}
 
-__global__ void k_cm_export(const double *__restrict__ cm, int64_t ldc,
-                            double *__restrict__ dst, int64_t ldd, int rows,
-                            int cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)rows * cols) return;
-    const int64_t r = i / cols, c = i % cols;   // write coalesced over c
-    dst[r * ldd + c] = cm[c * ldc + r];
-}
 
 // ---------------------------------------------------------------------------
 // panel column step: argmax + swap + scale + rank-1 on a col-major sub-panel
@@ -1098,26 +1065,8 @@ void launch_row_scatter(const double *src, int64_t lds, double *dst,
                        dim3(256), 0, s, src, lds, dst, ldd, idx, n_rows, cols);
 }
 
-void launch_laswp(double *A, int64_t lda, int i0, const int *piv, int nswap,
-                  int64_t c0, int64_t c1, hipStream_t s) {
-    if (nswap <= 0 || c1 <= c0) return;
-    hipLaunchKernelGGL(k_laswp, dim3(cdiv64(c1 - c0, 256)), dim3(256), 0, s, A,
-                       lda, i0, piv, nswap, c0, c1);
-}
 
-void launch_cm_import(const double *src, int64_t lds, double *cm, int64_t ldc,
-                      int rows, int cols, hipStream_t s) {
-    if (rows <= 0 || cols <= 0) return;
-    hipLaunchKernelGGL(k_cm_import, dim3(cdiv64((int64_t)rows * cols, 256)),
-                       dim3(256), 0, s, src, lds, cm, ldc, rows, cols);
-}
 
-void launch_cm_export(const double *cm, int64_t ldc, double *dst, int64_t ldd,
-                      int rows, int cols, hipStream_t s) {
-    if (rows <= 0 || cols <= 0) return;
-    hipLaunchKernelGGL(k_cm_export, dim3(cdiv64((int64_t)rows * cols, 256)),
-                       dim3(256), 0, s, cm, ldc, dst, ldd, rows, cols);
-}
 
 void launch_swap_map(const int *piv, int nb, int i0, int *dst_idx,
                      int *src_idx, hipStream_t s) {

@@ -21,12 +21,6 @@ void launch_row_gather(const double *src, int64_t lds, double *dst,
 void launch_row_scatter(const double *src, int64_t lds, double *dst,
                         int64_t ldd, const int *idx, int n_rows, int64_t cols,
                         hipStream_t s);
-void launch_laswp(double *A, int64_t lda, int i0, const int *piv, int nswap,
-                  int64_t c0, int64_t c1, hipStream_t s);
-void launch_cm_import(const double *src, int64_t lds, double *cm, int64_t ldc,
-                      int rows, int cols, hipStream_t s);
-void launch_cm_export(const double *cm, int64_t ldc, double *dst, int64_t ldd,
-                      int rows, int cols, hipStream_t s);
 void launch_swap_map(const int *piv, int nb, int i0, int *dst_idx,
                      int *src_idx, hipStream_t s);
 void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
