@@ -29,6 +29,9 @@ def _load():
         ctypes.c_char_p, ctypes.POINTER(ctypes.c_void_p)]
     lib.conflux_lu_make_uid.argtypes = [ctypes.c_char_p]
     lib.conflux_lu_init_matrix.argtypes = [ctypes.c_void_p, ctypes.c_uint64]
+    lib.conflux_lu_init_matrix_spd.argtypes = [ctypes.c_void_p, ctypes.c_uint64]
+    lib.conflux_chol_factor.argtypes = [ctypes.c_void_p,
+                                        ctypes.POINTER(ctypes.c_double)]
     lib.conflux_lu_set_matrix_local.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
     lib.conflux_lu_set_matrix_sim.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                               ctypes.c_void_p]
@@ -141,6 +144,14 @@ class Engine:
     def factor(self):
         ms = ctypes.c_double()
         _chk(lib().conflux_lu_factor(self._h, ctypes.byref(ms)), "factor")
+        return ms.value
+
+    def init_matrix_spd(self, seed=42):
+        _chk(lib().conflux_lu_init_matrix_spd(self._h, seed), "init_spd")
+
+    def factor_cholesky(self):
+        ms = ctypes.c_double()
+        _chk(lib().conflux_chol_factor(self._h, ctypes.byref(ms)), "chol")
         return ms.value
 
     def get_perm(self):

@@ -1,0 +1,140 @@
+// cholesky_miniapp — drop-in CLI for the reference CONFCHOX miniapp
+// (reference examples/cholesky_miniapp.cpp:63-90 flags: --dim --tile --grid
+//  --run).  Computes A = L L^T for the seeded synthetic SPD matrix on the
+// MI355X engine (conflux_chol_factor).
+//
+//   cholesky_miniapp --dim 65536 --tile 512 --grid 2,2,2 --run 5 [--sim]
+//
+// Multi-rank runs self-spawn one process per GPU (like conflux_miniapp);
+// --sim runs all grid ranks on one GPU.  Tile-size default heuristic
+// mirrors the reference (Cholesky.cpp:115-134).
+#include <hip/hip_runtime.h>
+#include <sys/wait.h>
+#include <unistd.h>
+
+#include <cmath>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include "../../include/conflux_lu.h"
+
+int main(int argc, char **argv) {
+    int N = 65536, v = 0, runs = 5, Px = 0, Py = 0, Pz = 0;
+    bool sim = false, timing = false;
+    for (int i = 1; i < argc; ++i) {
+        std::string a = argv[i];
+        auto val = [&]() -> std::string {
+            if (a.find('=') != std::string::npos) return a.substr(a.find('=') + 1);
+            return (i + 1 < argc) ? argv[++i] : "";
+        };
+        if (a == "-h" || a == "--help") {
+            std::printf("cholesky_miniapp --dim N --tile v --grid Px,Py,Pz "
+                        "--run r [--sim] [--timing]\n");
+            return 0;
+        } else if (a == "-N" || a.rfind("--dim", 0) == 0) N = std::atoi(val().c_str());
+        else if (a == "-v" || a.rfind("--tile", 0) == 0) v = std::atoi(val().c_str());
+        else if (a == "-r" || a.rfind("--run", 0) == 0) runs = std::atoi(val().c_str());
+        else if (a == "--sim") sim = true;
+        else if (a == "--timing") timing = true;
+        else if (a == "-g" || a.rfind("--grid", 0) == 0) {
+            std::string g = val();
+            if (std::sscanf(g.c_str(), "%d,%d,%d", &Px, &Py, &Pz) != 3) {
+                std::fprintf(stderr, "bad --grid\n");
+                return 1;
+            }
+        } else {
+            std::fprintf(stderr, "unknown arg %s\n", a.c_str());
+            return 1;
+        }
+    }
+    if (Px <= 0 || Py <= 0 || Pz <= 0) { Px = Py = Pz = 1; }
+    if (v == 0) {
+        // reference tile heuristic (Cholesky.cpp:115-134)
+        const double ratio = ((double)N * N * Pz / (Px * Py * Pz)) / 1e6;
+        v = ratio < 2.5 ? 128 : ratio < 30 ? 256 : ratio < 250 ? 512 : 1024;
+    }
+    const int P = Px * Py * Pz;
+    const int ntx = (N + v * Px - 1) / (v * Px);
+    N = v * Px * ntx;
+
+    int rank = sim ? -1 : 0, world = P;
+    char uid[CONFLUX_LU_UID_BYTES];
+    if (!sim && P > 1) {
+        if (const char *er = std::getenv("CONFLUX_RANK")) {
+            rank = std::atoi(er);
+            const char *uf = std::getenv("CONFLUX_UID_FILE");
+            FILE *f = uf ? std::fopen(uf, "rb") : nullptr;
+            if (!f || std::fread(uid, 1, sizeof uid, f) != sizeof uid) return 1;
+            std::fclose(f);
+        } else {
+            int ndev = 0;
+            (void)hipGetDeviceCount(&ndev);
+            if (ndev < P) {
+                std::fprintf(stderr, "%d GPUs for grid of %d (use --sim)\n",
+                             ndev, P);
+                return 1;
+            }
+            std::vector<int> pipes(2 * P);
+            for (int r = 0; r < P; ++r)
+                if (pipe(&pipes[2 * r])) return 1;
+            std::vector<pid_t> pids(P);
+            pid_t child = 0;
+            for (int r = 0; r < P; ++r) {
+                pid_t pid = fork();
+                if (pid == 0) {
+                    (void)hipSetDevice(r);
+                    if (r == 0) {
+                        if (conflux_lu_make_uid(uid)) _exit(1);
+                        for (int q = 1; q < P; ++q)
+                            if (write(pipes[2 * q + 1], uid, sizeof uid) !=
+                                (ssize_t)sizeof uid)
+                                _exit(1);
+                    } else if (read(pipes[2 * r], uid, sizeof uid) !=
+                               (ssize_t)sizeof uid) {
+                        _exit(1);
+                    }
+                    rank = r;
+                    child = 1;
+                    break;
+                }
+                pids[r] = pid;
+            }
+            if (!child) {
+                int status = 0, bad = 0;
+                for (int r = 0; r < P; ++r) {
+                    waitpid(pids[r], &status, 0);
+                    if (!WIFEXITED(status) || WEXITSTATUS(status)) bad = 1;
+                }
+                return bad;
+            }
+        }
+    }
+
+    conflux_lu_ctx *ctx = nullptr;
+    if (conflux_lu_create(N, v, Px, Py, Pz, rank, world,
+                          (world > 1 && !sim) ? uid : nullptr, &ctx)) {
+        std::fprintf(stderr, "create failed\n");
+        return 1;
+    }
+    conflux_lu_store_factors(ctx, timing ? 0 : 1);
+    const bool print0 = rank <= 0;
+    if (print0)
+        std::printf("cholesky (MI355X engine): N=%d v=%d grid %dx%dx%d\n", N,
+                    v, Px, Py, Pz);
+    for (int i = 0; i < runs + 1; ++i) {
+        conflux_lu_init_matrix_spd(ctx, 42);
+        double ms = 0;
+        if (conflux_chol_factor(ctx, &ms)) return 1;
+        if (i > 0 && print0) {
+            const double gflops = N / 3.0 * N / (ms * 1e-3) * N / 1e9;
+            std::printf("_result_ chol,conflux,%d,%d,%dx%dx%d,time,%.0f,%d "
+                        "(%.1f GFLOP/s)\n",
+                        N, P, Px, Py, Pz, ms, v, gflops);
+        }
+    }
+    conflux_lu_destroy(ctx);
+    return 0;
+}

@@ -269,7 +269,7 @@ int trsm_right_upper(Ctx &c, RankState &r, double *X, int64_t ldx, int M) {
     for (int jb = 0; jb < v; jb += NB) {
         const int nb = std::min(NB, v - jb);
         launch_trsm_right_upper32(r.A00 + i64(jb) * v + jb, v, X + jb, ldx, nb,
-                                  M, c.stream);
+                                  M, 0, c.stream);
         if (jb + nb < v)
             launch_dgemm_f64(X + jb, ldx, r.A00 + i64(jb) * v + jb + nb, v,
                              X + jb + nb, ldx, M, v - jb - nb, nb, c.stream);
@@ -1048,6 +1048,336 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
     return 0;
 }
 
+
+// ===========================================================================
+// Cholesky (CONFCHOX path — SURVEY §8f1, reference src/conflux/cholesky/):
+// the same tile-cyclic (Px,Py,Pz) machinery minus pivoting.  Per tile-column
+// k (Cholesky.cpp flow: choleskyA00 :192, updateA10 :280, computeA11
+// :345-351, reduce :581-620):
+//   c0  depth-reduce the k-th tile column to layer 0
+//   c1  potrf the diagonal tile on its owner; broadcast L_kk down the column
+//   c2  L_ik = A_ik * L_kk^-T on the column ranks (strictly below diagonal)
+//   c2b slab-spread the L panel over (pj, pk)            [= LU C8]
+//   c3  transpose-spread: every rank receives L_jk slabs for its local
+//       column tiles j (the reference's A01rcv representatives)
+//   c4  A_ij -= L_ik * L_jk^T for local tiles with i >= j > k (NT GEMM)
+// No pivoting -> row activation is static: rows of tiles < t owned by pi.
+// ===========================================================================
+namespace {
+
+inline int ntiles_lt(const Ctx &c, int pi, int t) {
+    return (t <= pi) ? 0 : (t - pi + c.Px - 1) / c.Px;
+}
+
+int potrf_tile(Ctx &c, double *T, int64_t ld) {
+    const int v = c.v, NB = conflux_panel_nb();
+    size_t slot;
+    if (ev_begin(c, 1, 0, &slot)) return CONFLUX_LU_EHIP;
+    for (int jb = 0; jb < v; jb += NB) {
+        const int nb = std::min(NB, v - jb);
+        launch_potrf32(T + i64(jb) * ld + jb, ld, nb, c.stream);
+        if (jb + nb < v) {
+            launch_trsm_right_upper32(T + i64(jb) * ld + jb, ld,
+                                      T + i64(jb + nb) * ld + jb, ld, nb,
+                                      v - jb - nb, /*trans=*/1, c.stream);
+            launch_dgemm_f64_nt(T + i64(jb + nb) * ld + jb, ld,
+                                T + i64(jb + nb) * ld + jb, ld,
+                                T + i64(jb + nb) * ld + jb + nb, ld,
+                                v - jb - nb, v - jb - nb, nb, c.stream);
+        }
+    }
+    return ev_end(c, slot);
+}
+
+int trsm_right_lowT(Ctx &c, RankState &r, double *X, int64_t ldx, int M) {
+    const int v = c.v, NB = conflux_panel_nb();
+    size_t slot;
+    if (ev_begin(c, 2, 0, &slot)) return CONFLUX_LU_EHIP;
+    for (int jb = 0; jb < v; jb += NB) {
+        const int nb = std::min(NB, v - jb);
+        launch_trsm_right_upper32(r.A00 + i64(jb) * v + jb, v, X + jb, ldx, nb,
+                                  M, /*trans=*/1, c.stream);
+        if (jb + nb < v)
+            launch_dgemm_f64_nt(X + jb, ldx, r.A00 + i64(jb + nb) * v + jb, v,
+                                X + jb + nb, ldx, M, v - jb - nb, nb,
+                                c.stream);
+    }
+    return ev_end(c, slot);
+}
+
+int chol_step(Ctx &c, int k) {
+    const int v = c.v, Px = c.Px, Py = c.Py, Pz = c.Pz, Nt = c.Nt;
+    const int64_t Nl = c.Nl;
+    const int kcol = k % Py, krow = k % Px;
+    const int loff = (k / Py) * v;
+    auto fnp_of = [&](int pi) { return v * ntiles_lt(c, pi, k); };
+    auto f2_of = [&](int pi) { return fnp_of(pi) + (pi == krow ? v : 0); };
+
+    // ---- c0: copy the k-th tile column into A10, depth-reduce -------------
+    for (auto &r : c.rs) {
+        if (r.pj != kcol) continue;
+        const int f = fnp_of(r.pi);
+        launch_copy2d(r.A11 + i64(f) * Nl + loff, Nl, r.A10 + i64(f) * v, v,
+                      c.Ml - f, v, c.stream);
+    }
+    if (Pz > 1) {
+        if (!c.sim) NCCLCHK(ncclGroupStart());
+        for (int pi = 0; pi < Px; ++pi) {
+            if (!c.sim && (c.rs[0].pi != pi || c.rs[0].pj != kcol)) continue;
+            const int f = fnp_of(pi);
+            auto buf = [&, f](RankState &x) { return x.A10 + i64(f) * v; };
+            if (reduce_over_pk(c, pi, kcol, i64(c.Ml - f) * v, buf))
+                return CONFLUX_LU_ECOMM;
+        }
+        if (!c.sim) {
+            NCCLCHK(ncclGroupEnd());
+            RankState &me = c.rs[0];
+            if (me.pj == kcol) {
+                const int f = fnp_of(me.pi);
+                auto buf = [&, f](RankState &x) { return x.A10 + i64(f) * v; };
+                if (reduce_over_pk_finish(c, me.pi, kcol, i64(c.Ml - f) * v,
+                                          buf))
+                    return CONFLUX_LU_ECOMM;
+            }
+        }
+    }
+
+    // ---- c1: potrf diagonal tile; broadcast L_kk down the column ----------
+    {
+        RankState *own = get_rs(c, krow, kcol, 0);
+        if (own) {
+            const int fd = fnp_of(krow);
+            if (potrf_tile(c, own->A10 + i64(fd) * v, v))
+                return CONFLUX_LU_EINTERNAL;
+            launch_copy2d(own->A10 + i64(fd) * v, v, own->A00, v, v, v,
+                          c.stream);
+        }
+        if (c.sim) {
+            for (int pi = 0; pi < Px; ++pi) {
+                if (pi == krow) continue;
+                RankState &d = *get_rs(c, pi, kcol, 0);
+                if (d2d(c, d.A00, own->A00, i64(v) * v)) return CONFLUX_LU_EHIP;
+            }
+        } else {
+            RankState &me = c.rs[0];
+            if (me.pj == kcol && me.pk == 0 && Px > 1) {
+                NCCLCHK(ncclGroupStart());
+                if (me.pi == krow) {
+                    for (int pi = 0; pi < Px; ++pi)
+                        if (pi != krow)
+                            NCCLCHK(ncclSend(me.A00, i64(v) * v, ncclDouble,
+                                             grank_of(c, pi, kcol, 0), c.comm,
+                                             c.stream));
+                } else {
+                    NCCLCHK(ncclRecv(me.A00, i64(v) * v, ncclDouble,
+                                     grank_of(c, krow, kcol, 0), c.comm,
+                                     c.stream));
+                }
+                NCCLCHK(ncclGroupEnd());
+            }
+        }
+    }
+
+    // ---- c2: column TRSM (strictly below the diagonal); store L -----------
+    for (auto &r : c.rs) {
+        if (r.pj != kcol || r.pk != 0) continue;
+        const int f2 = f2_of(r.pi);
+        const int n2 = c.Ml - f2;
+        if (n2 > 0 &&
+            trsm_right_lowT(c, r, r.A10 + i64(f2) * v, v, n2))
+            return CONFLUX_LU_EINTERNAL;
+        if (c.store_factors) {
+            if (n2 > 0)
+                launch_copy2d(r.A10 + i64(f2) * v, v,
+                              r.Fres + i64(f2) * Nl + loff, Nl, n2, v,
+                              c.stream);
+            if (r.pi == krow)  // diagonal tile: L_kk (upper half junk;
+                               // consumers read tril only)
+                launch_copy2d(r.A00, v, r.Fres + i64(fnp_of(krow)) * Nl + loff,
+                              Nl, v, v, c.stream);
+        }
+    }
+
+    // ---- c2b: slab-spread the L panel over (pj, pk)  [= LU C8] ------------
+    if (Py == 1 && Pz == 1) {
+        for (auto &r : c.rs) {
+            const int f2 = f2_of(r.pi);
+            launch_copy2d(r.A10 + i64(f2) * v, v, r.A10Rcv, c.nlayr,
+                          c.Ml - f2, v, c.stream);
+        }
+    } else {
+        for (int pi = 0; pi < Px; ++pi) {
+            RankState *root = get_rs(c, pi, kcol, 0);
+            const int f2 = f2_of(pi);
+            const int n2 = c.Ml - f2;
+            if (c.sim) {
+                if (n2 > 0)
+                    launch_slab_pack(root->A10 + i64(f2) * v, v, n2, c.nlayr,
+                                     Pz, root->slabs, c.stream);
+                for (int pj = 0; pj < Py; ++pj)
+                    for (int pk = 0; pk < Pz; ++pk) {
+                        RankState &d = *get_rs(c, pi, pj, pk);
+                        if (n2 > 0 &&
+                            d2d(c, d.A10Rcv, root->slabs + i64(pk) * n2 * c.nlayr,
+                                i64(n2) * c.nlayr))
+                            return CONFLUX_LU_EHIP;
+                    }
+            } else {
+                RankState &me = c.rs[0];
+                if (me.pi != pi || n2 <= 0) continue;
+                NCCLCHK(ncclGroupStart());
+                if (root && root->grank == me.grank) {
+                    launch_slab_pack(me.A10 + i64(f2) * v, v, n2, c.nlayr, Pz,
+                                     me.slabs, c.stream);
+                    for (int pj = 0; pj < Py; ++pj)
+                        for (int pk = 0; pk < Pz; ++pk) {
+                            if (pj == kcol && pk == 0) continue;
+                            NCCLCHK(ncclSend(me.slabs + i64(pk) * n2 * c.nlayr,
+                                             i64(n2) * c.nlayr, ncclDouble,
+                                             grank_of(c, pi, pj, pk), c.comm,
+                                             c.stream));
+                        }
+                } else {
+                    NCCLCHK(ncclRecv(me.A10Rcv, i64(n2) * c.nlayr, ncclDouble,
+                                     grank_of(c, pi, kcol, 0), c.comm,
+                                     c.stream));
+                }
+                NCCLCHK(ncclGroupEnd());
+                if (root && root->grank == me.grank)
+                    if (d2d(c, me.A10Rcv, me.slabs, i64(n2) * c.nlayr))
+                        return CONFLUX_LU_EHIP;
+            }
+        }
+    }
+
+    // ---- c3: transpose-spread L_jk slabs to the columns that need them ----
+    // destination layout: A01Rcv as [local col tile][v][nlayr]
+    for (int j = k + 1; j < Nt; ++j) {
+        const int rpi = j % Px, dpj = j % Py;
+        const int f2r = f2_of(rpi);
+        const int rowoff = ntiles_lt(c, rpi, j) * v - f2r;  // rows into slab
+        const int ltj = j / Py;
+        RankState *root = get_rs(c, rpi, kcol, 0);
+        if (c.sim) {
+            for (int pi = 0; pi < Px; ++pi)
+                for (int pk = 0; pk < Pz; ++pk) {
+                    RankState &d = *get_rs(c, pi, dpj, pk);
+                    const int n2 = c.Ml - f2r;
+                    const double *srcslab = (Py == 1 && Pz == 1)
+                        ? root->A10Rcv + i64(rowoff) * c.nlayr
+                        : root->slabs + (i64(pk) * n2 + rowoff) * c.nlayr;
+                    if (d2d(c, d.A01Rcv + i64(ltj) * v * c.nlayr, srcslab,
+                            i64(v) * c.nlayr))
+                        return CONFLUX_LU_EHIP;
+                }
+        } else {
+            RankState &me = c.rs[0];
+            const bool is_root = root && root->grank == me.grank;
+            const bool is_dst = (me.pj == dpj);
+            const int n2 = c.Ml - f2r;
+            if (is_root) {
+                const double *base = (Py == 1 && Pz == 1) ? me.A10Rcv : me.slabs;
+                for (int pi = 0; pi < Px; ++pi)
+                    for (int pk = 0; pk < Pz; ++pk) {
+                        const double *srcslab = (Py == 1 && Pz == 1)
+                            ? base + i64(rowoff) * c.nlayr
+                            : base + (i64(pk) * n2 + rowoff) * c.nlayr;
+                        if (pi == me.pi && dpj == me.pj && pk == me.pk) {
+                            if (d2d(c, me.A01Rcv + i64(ltj) * v * c.nlayr,
+                                    srcslab, i64(v) * c.nlayr))
+                                return CONFLUX_LU_EHIP;
+                        } else {
+                            NCCLCHK(ncclGroupStart());
+                            NCCLCHK(ncclSend(srcslab, i64(v) * c.nlayr,
+                                             ncclDouble,
+                                             grank_of(c, pi, dpj, pk), c.comm,
+                                             c.stream));
+                            NCCLCHK(ncclGroupEnd());
+                        }
+                    }
+            } else if (is_dst) {
+                NCCLCHK(ncclGroupStart());
+                NCCLCHK(ncclRecv(me.A01Rcv + i64(ltj) * v * c.nlayr,
+                                 i64(v) * c.nlayr, ncclDouble,
+                                 grank_of(c, rpi, kcol, 0), c.comm, c.stream));
+                NCCLCHK(ncclGroupEnd());
+            }
+        }
+    }
+
+    // ---- c4: low-rank updates of local tiles with i >= j > k --------------
+    for (auto &r : c.rs) {
+        const int f2 = f2_of(r.pi);
+        for (int ltj = 0; ltj < c.tA11y; ++ltj) {
+            const int gtj = ltj * Py + r.pj;
+            if (gtj <= k || gtj >= Nt) continue;
+            const int rstart = v * ntiles_lt(c, r.pi, gtj);
+            const int M2 = c.Ml - rstart;
+            if (M2 <= 0) continue;
+            const double fl = 2.0 * M2 * (double)v * c.nlayr;
+            size_t slot;
+            if (ev_begin(c, 0, fl, &slot)) return CONFLUX_LU_EHIP;
+            launch_dgemm_f64_nt(r.A10Rcv + i64(rstart - f2) * c.nlayr, c.nlayr,
+                                r.A01Rcv + i64(ltj) * v * c.nlayr, c.nlayr,
+                                r.A11 + i64(rstart) * Nl + i64(ltj) * v, Nl,
+                                M2, v, c.nlayr, c.stream);
+            if (ev_end(c, slot)) return CONFLUX_LU_EHIP;
+        }
+    }
+    return 0;
+}
+
+int chol_loop(Ctx &c, double *elapsed_ms) {
+    for (auto &r : c.rs) {
+        if (c.store_factors) {
+            if (ensure_factor_bufs(c, r)) return CONFLUX_LU_EHIP;
+            launch_zero2d(r.Fres, c.Nl, c.Ml, c.Nl, c.stream);
+        }
+    }
+    // perm is identity for the pivotless path (API consistency)
+    c.pivotInds.resize(c.M);
+    std::iota(c.pivotInds.begin(), c.pivotInds.end(), 0);
+    c.evs_used = 0;
+    for (auto &t : c.cats) t = TimeCat{};
+
+    HIPCHK(hipStreamSynchronize(c.stream));
+    if (c.have_comm) {
+        static double *dummy = nullptr;
+        if (!dummy) HIPCHK(hipMalloc(&dummy, 8));
+        NCCLCHK(ncclAllReduce(dummy, dummy, 1, ncclDouble, ncclSum, c.comm,
+                              c.stream));
+        HIPCHK(hipStreamSynchronize(c.stream));
+    }
+    const auto t1 = std::chrono::high_resolution_clock::now();
+    for (int k = 0; k < c.Nt; ++k) {
+        int rc = chol_step(c, k);
+        if (rc) return rc;
+    }
+    HIPCHK(hipStreamSynchronize(c.stream));
+    if (c.have_comm) {
+        static double *dummy2 = nullptr;
+        if (!dummy2) HIPCHK(hipMalloc(&dummy2, 8));
+        NCCLCHK(ncclAllReduce(dummy2, dummy2, 1, ncclDouble, ncclSum, c.comm,
+                              c.stream));
+        HIPCHK(hipStreamSynchronize(c.stream));
+    }
+    const auto t2 = std::chrono::high_resolution_clock::now();
+    if (elapsed_ms)
+        *elapsed_ms =
+            std::chrono::duration<double, std::milli>(t2 - t1).count();
+    for (size_t i = 0; i < c.evs_used; ++i) {
+        float ms = 0;
+        HIPCHK(hipEventElapsedTime(&ms, c.evs[i].a, c.evs[i].b));
+        TimeCat &t = c.cats[c.evs[i].cat];
+        t.seconds += ms * 1e-3;
+        t.launches += 1;
+        t.flops += c.evs[i].flops;
+    }
+    return 0;
+}
+
+}  // namespace
+
 }  // namespace
 
 // ===========================================================================
@@ -1145,6 +1475,27 @@ int conflux_lu_init_matrix(conflux_lu_ctx *c, uint64_t seed) {
                            r.pk != 0, seed, c->stream);
     HIPCHK(hipStreamSynchronize(c->stream));
     return CONFLUX_LU_OK;
+}
+
+/* SPD fill for the Cholesky path: sym(gen) + 2N on the diagonal */
+int conflux_lu_init_matrix_spd(conflux_lu_ctx *c, uint64_t seed) {
+    for (auto &r : c->rs)
+        launch_init_matrix_spd(r.A11, c->Ml, c->Nl, c->v, c->Px, c->Py, r.pi,
+                               r.pj, r.pk != 0, seed, c->N, c->stream);
+    HIPCHK(hipStreamSynchronize(c->stream));
+    return CONFLUX_LU_OK;
+}
+
+/* Cholesky factorization A = L L^T of the current (SPD) matrix; the
+ * CONFCHOX path (reference src/conflux/cholesky/Cholesky.cpp:857
+ * parallelCholesky).  Fres then holds L in the tile-cyclic layout (lower
+ * triangle valid). */
+int conflux_chol_factor(conflux_lu_ctx *c, double *elapsed_ms) {
+    int rc = chol_loop(*c, elapsed_ms);
+    if (rc)
+        std::fprintf(stderr, "[conflux_lu] chol failed: %s\n",
+                     c->err.c_str());
+    return rc;
 }
 
 int conflux_lu_set_matrix_local(conflux_lu_ctx *c, const double *local) {

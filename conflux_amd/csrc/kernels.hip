@@ -47,9 +47,16 @@ DEVFN uint64_t splitmix64(uint64_t z) {
     return z;
 }
 
+DEVFN double gen_entry(int64_t gi, int64_t gj, uint64_t seed) {
+    uint64_t key = ((uint64_t)gi << 32) ^ (uint64_t)gj;
+    key += seed * 0xBF58476D1CE4E5B9ull;
+    const uint64_t h = splitmix64(key);
+    return 5.0 + (double)(h >> 11) * (1.0 / 9007199254740992.0);
+}
+
 __global__ void k_init_matrix(double *__restrict__ A, int Ml, int Nl, int v,
                               int Px, int Py, int pi, int pj, int zero_layer,
-                              uint64_t seed) {
+                              uint64_t seed, int spd, int64_t Nglob) {
     const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     const int64_t total = (int64_t)Ml * Nl;
     if (idx >= total) return;
@@ -58,10 +65,14 @@ __global__ void k_init_matrix(double *__restrict__ A, int Ml, int Nl, int v,
     // local (r, c) -> global (i, j): tile-cyclic map (layout.cpp:95-123)
     const int64_t gi = (int64_t)(r / v * Px + pi) * v + r % v;
     const int64_t gj = (int64_t)(c / v * Py + pj) * v + c % v;
-    uint64_t key = ((uint64_t)gi << 32) ^ (uint64_t)gj;
-    key += seed * 0xBF58476D1CE4E5B9ull;
-    const uint64_t h = splitmix64(key);
-    A[idx] = 5.0 + (double)(h >> 11) * (1.0 / 9007199254740992.0);
+    if (spd) {
+        // symmetric positive definite: sym(gen)/1 + 2N on the diagonal
+        double x = 0.5 * (gen_entry(gi, gj, seed) + gen_entry(gj, gi, seed));
+        if (gi == gj) x += 2.0 * (double)Nglob;
+        A[idx] = x;
+        return;
+    }
+    A[idx] = gen_entry(gi, gj, seed);
 }
 
 // ---------------------------------------------------------------------------
@@ -540,11 +551,15 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
 // row r.  cblas_dtrsm Right/Upper/NoTrans/NonUnit (conflux_opt.hpp:1347).
 __global__ __launch_bounds__(256) void k_trsm_right_upper32(
     const double *__restrict__ U, int64_t ldu, double *__restrict__ X,
-    int64_t ldx, int nb, int64_t M) {
+    int64_t ldx, int nb, int64_t M, int trans) {
     __shared__ double sU[32][33];
     __shared__ double sX[256][33];
     const int tid = threadIdx.x;
-    for (int i = tid; i < nb * nb; i += 256) sU[i / nb][i % nb] = U[(i / nb) * ldu + i % nb];
+    // trans: sU[i][c] = U[c][i] — solves X*L^T = B for lower-triangular L
+    // (cblas_dtrsm Right/Lower/Trans/NonUnit, reference Cholesky.cpp:280)
+    for (int i = tid; i < nb * nb; i += 256)
+        sU[i / nb][i % nb] = trans ? U[(i % nb) * ldu + i / nb]
+                                   : U[(i / nb) * ldu + i % nb];
     const int64_t r0 = (int64_t)blockIdx.x * 256;
     const int rows = (int)min((int64_t)256, M - r0);
     // stage rows r0..r0+rows coalesced: thread t covers elements t, t+256, ...
@@ -834,6 +849,137 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
 }
 
 
+// NT variant: C -= A * B^T with B stored (N x K) row-major — the
+// computeA11 low-rank update (reference Cholesky.cpp:345-351).
+__global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8_nt(
+    const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
+    int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
+    int ntm, int ntn) {
+    // bijective XCD swizzle (guide §5: q/r form)
+    int wg = blockIdx.x;
+    {
+        const int nwg = ntm * ntn;
+        const int q = nwg >> 3, r = nwg & 7;
+        const int xcd = wg & 7, idx = wg >> 3;
+        // inverse of dispatch round-robin: give each XCD a contiguous chunk
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+        if (nwg < 8) wg = blockIdx.x;   // tiny grids: identity
+    }
+    const int tm = wg / ntn, tn = wg % ntn;
+    const int row0 = tm * GEMM_BM;
+    const int64_t col0 = (int64_t)tn * GEMM_BN;
+
+    __shared__ double As[2][GEMM_BK][GEMM_BM + 1];   // transposed, padded
+    __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];   // double-buffered
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm0 = (wave >> 2) * 64;             // wave's 64x32 sub-tile
+    const int wn0 = (wave & 3) * 32;
+    const int frow = lane & 15;                   // fragment row/col lane part
+    const int fk = lane >> 4;                     // fragment k lane part
+
+    f64x4 acc[4][2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = f64x4{0, 0, 0, 0};
+
+    // staging registers: 8 A elements + 8 B elements per thread per K-tile
+    // (element e = tid + i*256 over the 2048-element tile; A reads coalesce
+    // over the 16-wide rows, B reads over the 128-wide rows)
+    double ra[4], rb[4];
+    const int ktiles = (K + GEMM_BK - 1) / GEMM_BK;
+
+    auto load_a = [&](int kt) {
+        const int kk = kt * GEMM_BK;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;               // 0..2047
+            const int r = e >> 4, k = e & 15;          // row-major in tile
+            const int gr = row0 + r;
+            ra[i] = (gr < M && kk + k < K) ? A[(int64_t)gr * lda + kk + k] : 0.0;
+        }
+    };
+    auto load_b = [&](int kt) {
+        // B is (N x K) row-major here: Bs[k][n] <- B[col0+n][kk+k]
+        // (coalesced over k within each B row, like the A staging)
+        const int kk = kt * GEMM_BK;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;               // 0..2047
+            const int n = e >> 4, k = e & 15;
+            const int64_t gc = col0 + n;
+            rb[i] = (kk + k < K && gc < N) ? B[gc * ldb + kk + k] : 0.0;
+        }
+    };
+    auto write_lds = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;
+            As[buf][e & 15][e >> 4] = ra[i];
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;
+            Bs[buf][e & 15][e >> 4] = rb[i];
+        }
+    };
+
+    load_a(0);
+    load_b(0);
+    write_lds(0);
+    __syncthreads();
+
+    int cur = 0;
+    for (int kt = 0; kt < ktiles; ++kt) {
+        if (kt + 1 < ktiles) {           // issue next tile's global loads
+            load_a(kt + 1);
+            load_b(kt + 1);
+        }
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            const int k = kk * 4 + fk;
+            double af[4], bf[2];
+#pragma unroll
+            for (int i = 0; i < 4; ++i) af[i] = As[cur][k][wm0 + i * 16 + frow];
+#pragma unroll
+            for (int j = 0; j < 2; ++j) bf[j] = Bs[cur][k][wn0 + j * 16 + frow];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 2; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+        // write t+1 into the other buffer (everyone finished reading it at
+        // the barrier that ended step t-1), then one barrier per K-step
+        if (kt + 1 < ktiles) write_lds(cur ^ 1);
+        __syncthreads();
+        cur ^= 1;
+    }
+
+    // epilogue: C -= acc   (read-modify-write, coalesced over frag columns)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                // f64 16x16x4 C/D map (hardware-verified by mfma_probe):
+                // lane l, reg q -> D[4*q + (l>>4)][l & 15]
+                const int r = row0 + wm0 + i * 16 + q * 4 + fk;
+                const int64_t cidx = col0 + wn0 + j * 16 + frow;
+                if (r < M && cidx < N) {
+                    C[(int64_t)r * ldc + cidx] -= acc[i][j][q];
+                }
+            }
+        }
+    }
+}
+
+
 // ---------------------------------------------------------------------------
 // glds variant: async global->LDS staging (__builtin_amdgcn_global_load_lds,
 // 16-byte pieces), one barrier per K-step, XOR-swizzled SOURCE addresses so
@@ -973,6 +1119,41 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_glds(
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// Cholesky kernels (CONFCHOX path; reference
+// src/conflux/cholesky/Cholesky.cpp): k_potrf32 <- the 32-wide micro-factor
+// of LAPACKE_dpotrf('L') (Cholesky.cpp:192); the right/lower/trans TRSM is
+// k_trsm_right_tri32 with trans=1 (updateA10's dtrsm, Cholesky.cpp:280);
+// the NoTrans x Trans low-rank update (computeA11, Cholesky.cpp:345-351) is
+// k_dgemm_f64_w8_nt below.
+// ---------------------------------------------------------------------------
+// single block: in-place lower Cholesky of the nb x nb (nb <= 32) diagonal
+// block, row-major ld.  Sequential over columns in LDS (no pivoting).
+__global__ __launch_bounds__(256) void k_potrf32(double *__restrict__ A,
+                                                 int64_t lda, int nb) {
+    __shared__ double sA[32][33];
+    const int tid = threadIdx.x;
+    for (int i = tid; i < nb * nb; i += 256)
+        sA[i / nb][i % nb] = A[(i / nb) * lda + i % nb];
+    __syncthreads();
+    for (int c = 0; c < nb; ++c) {
+        if (tid == 0) sA[c][c] = sqrt(sA[c][c]);
+        __syncthreads();
+        const double inv = 1.0 / sA[c][c];
+        if (tid > c && tid < nb) sA[tid][c] *= inv;
+        __syncthreads();
+        // trailing update: element (i, j), i > c, c < j <= i
+        for (int e = tid; e < nb * nb; e += 256) {
+            const int i = e / nb, j = e % nb;
+            if (i > c && j > c && j <= i) sA[i][j] -= sA[i][c] * sA[j][c];
+        }
+        __syncthreads();
+    }
+    for (int i = tid; i < nb * nb; i += 256)
+        A[(i / nb) * lda + i % nb] = sA[i / nb][i % nb];
+}
+
 // ---------------------------------------------------------------------------
 // misc small kernels for the distributed path
 // ---------------------------------------------------------------------------
@@ -1025,7 +1206,17 @@ void launch_init_matrix(double *A, int Ml, int Nl, int v, int Px, int Py,
                         hipStream_t s) {
     const int64_t total = (int64_t)Ml * Nl;
     hipLaunchKernelGGL(k_init_matrix, dim3(cdiv64(total, 256)), dim3(256), 0, s,
-                       A, Ml, Nl, v, Px, Py, pi, pj, zero_layer, seed);
+                       A, Ml, Nl, v, Px, Py, pi, pj, zero_layer, seed, 0,
+                       (int64_t)0);
+}
+
+void launch_init_matrix_spd(double *A, int Ml, int Nl, int v, int Px, int Py,
+                            int pi, int pj, int zero_layer, uint64_t seed,
+                            int64_t Nglob, hipStream_t s) {
+    const int64_t total = (int64_t)Ml * Nl;
+    hipLaunchKernelGGL(k_init_matrix, dim3(cdiv64(total, 256)), dim3(256), 0, s,
+                       A, Ml, Nl, v, Px, Py, pi, pj, zero_layer, seed, 1,
+                       Nglob);
 }
 
 void launch_copy2d(const double *src, int64_t lds, double *dst, int64_t ldd,
@@ -1111,10 +1302,16 @@ void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
 }
 
 void launch_trsm_right_upper32(const double *U, int64_t ldu, double *X,
-                               int64_t ldx, int nb, int64_t M, hipStream_t s) {
+                               int64_t ldx, int nb, int64_t M, int trans,
+                               hipStream_t s) {
     if (M <= 0 || nb <= 0) return;
     hipLaunchKernelGGL(k_trsm_right_upper32, dim3(cdiv64(M, 256)), dim3(256),
-                       0, s, U, ldu, X, ldx, nb, M);
+                       0, s, U, ldu, X, ldx, nb, M, trans);
+}
+
+void launch_potrf32(double *A, int64_t lda, int nb, hipStream_t s) {
+    if (nb <= 0) return;
+    hipLaunchKernelGGL(k_potrf32, dim3(1), dim3(256), 0, s, A, lda, nb);
 }
 
 int g_dgemm_variant = -1;  // 0 = 4-wave, 1 = 8-wave; env CONFLUX_GEMM_VARIANT
@@ -1138,6 +1335,16 @@ void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
     else
         hipLaunchKernelGGL(k_dgemm_f64, dim3(ntm * ntn), dim3(GEMM_TPB), 0, s,
                            A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
+}
+
+void launch_dgemm_f64_nt(const double *A, int64_t lda, const double *B,
+                         int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
+                         int K, hipStream_t s) {
+    if (M <= 0 || N <= 0 || K <= 0) return;
+    const int ntm = (int)cdiv64(M, GEMM_BM);
+    const int ntn = (int)cdiv64(N, GEMM_BN);
+    hipLaunchKernelGGL(k_dgemm_f64_w8_nt, dim3(ntm * ntn), dim3(512), 0, s, A,
+                       lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
 }
 
 void launch_pack_candidate(const double *A10, int64_t lda, const int *gri,

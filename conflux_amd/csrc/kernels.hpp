@@ -35,7 +35,15 @@ void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
                                    int64_t ldx, int nb, int64_t N,
                                    hipStream_t s);
 void launch_trsm_right_upper32(const double *U, int64_t ldu, double *X,
-                               int64_t ldx, int nb, int64_t M, hipStream_t s);
+                               int64_t ldx, int nb, int64_t M, int trans,
+                               hipStream_t s);
+void launch_potrf32(double *A, int64_t lda, int nb, hipStream_t s);
+void launch_dgemm_f64_nt(const double *A, int64_t lda, const double *B,
+                         int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
+                         int K, hipStream_t s);
+void launch_init_matrix_spd(double *A, int Ml, int Nl, int v, int Px, int Py,
+                            int pi, int pj, int zero_layer, uint64_t seed,
+                            int64_t Nglob, hipStream_t s);
 void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
                       int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
                       int K, hipStream_t s);

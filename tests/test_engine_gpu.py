@@ -171,3 +171,65 @@ def test_lu_single_rank_medium(eng, N, v):
     for i, j in enumerate(piv):
         ref[i], ref[j] = ref[j], ref[i]
     assert np.array_equal(perm, ref)
+
+
+# ---------------- Cholesky (CONFCHOX path) ---------------------
+
+def _spd(N):
+    G = gen_matrix(N)
+    return 0.5 * (G + G.T) + 2.0 * N * np.eye(N)
+
+
+CHOL_GRIDS = [
+    (64, 8, 1, 1, 1),
+    (128, 16, 1, 1, 2),
+    (128, 16, 2, 2, 1),
+    (128, 16, 2, 2, 2),
+    (128, 8, 4, 4, 2),
+    (256, 32, 2, 2, 2),
+]
+
+
+@pytest.mark.parametrize("N,v,Px,Py,Pz", CHOL_GRIDS)
+def test_cholesky_parity(eng, N, v, Px, Py, Pz):
+    """Engine Cholesky vs scipy/LAPACK dpotrf on the same SPD input.
+    Pivotless path: no discrete choices, so LAPACK is the oracle."""
+    import scipy.linalg as la
+    A = _spd(N)
+    with eng.Engine(N, v, Px, Py, Pz, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor_cholesky()
+        F = e.get_F_global()
+    L = np.tril(F)
+    Lref = la.cholesky(A, lower=True)
+    assert np.abs(L - Lref).max() < 1e-9 * N
+    res = np.linalg.norm(A - L @ L.T) / np.linalg.norm(A)
+    assert res < 1e-14
+
+
+def test_cholesky_device_spd_generator(eng):
+    """init_matrix_spd (device) must realize the same SPD matrix as the
+    host formula: factor it and compare L against scipy of the host A."""
+    import scipy.linalg as la
+    N, v = 128, 16
+    with eng.Engine(N, v, 1, 1, 1, rank=-1) as e:
+        e.store_factors(True)
+        import conflux_amd, ctypes
+        conflux_amd.lib().conflux_lu_init_matrix_spd(e._h, 42)
+        e.factor_cholesky()
+        F = e.get_F_global()
+    Lref = la.cholesky(_spd(N), lower=True)
+    assert np.abs(np.tril(F) - Lref).max() < 1e-10
+
+
+def test_cholesky_residual_medium(eng):
+    N, v = 1024, 128
+    A = _spd(N)
+    with eng.Engine(N, v, 1, 1, 1, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor_cholesky()
+        F = e.get_F_global()
+    L = np.tril(F)
+    assert np.linalg.norm(A - L @ L.T) / np.linalg.norm(A) < 1e-14
