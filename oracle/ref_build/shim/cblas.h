@@ -3,6 +3,9 @@
  * Only the symbols conflux_opt.hpp uses: cblas_dgemm (:1628), cblas_dtrsm
  * (:1347, :1539).  Enum values are the standard CBLAS ABI constants. */
 #pragma once
+/* the reference's non-MKL branch includes only <cblas.h> but also calls
+ * LAPACKE_* (it was evidently built with MKL upstream) — chain the shim */
+#include "lapacke.h"
 #ifdef __cplusplus
 extern "C" {
 #endif
@@ -17,6 +20,10 @@ typedef CBLAS_LAYOUT CBLAS_ORDER;
 void cblas_dgemm(CBLAS_LAYOUT layout, CBLAS_TRANSPOSE TransA,
                  CBLAS_TRANSPOSE TransB, int M, int N, int K, double alpha,
                  const double *A, int lda, const double *B, int ldb,
+                 double beta, double *C, int ldc);
+
+void cblas_dsyrk(CBLAS_LAYOUT layout, CBLAS_UPLO Uplo, CBLAS_TRANSPOSE Trans,
+                 int N, int K, double alpha, const double *A, int lda,
                  double beta, double *C, int ldc);
 
 void cblas_dtrsm(CBLAS_LAYOUT layout, CBLAS_SIDE Side, CBLAS_UPLO Uplo,

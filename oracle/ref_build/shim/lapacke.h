@@ -9,6 +9,9 @@ extern "C" {
 #define LAPACK_COL_MAJOR 102
 typedef int lapack_int;
 
+lapack_int LAPACKE_dpotrf(int matrix_layout, char uplo, lapack_int n,
+                          double *a, lapack_int lda);
+
 lapack_int LAPACKE_dgetrf(int matrix_layout, lapack_int m, lapack_int n,
                           double *a, lapack_int lda, lapack_int *ipiv);
 
