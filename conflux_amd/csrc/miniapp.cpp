@@ -75,7 +75,6 @@ int main(int argc, char **argv) {
     // launch topology: rank from env (external launcher), else self-spawn
     int rank = -2, world = P;
     char uid[CONFLUX_LU_UID_BYTES];
-    std::vector<int> kids;
     if (sim || P == 1) {
         rank = sim ? -1 : 0;
         world = P;
@@ -93,7 +92,6 @@ int main(int argc, char **argv) {
         }
     }
     // (rank == -2 falls through to the self-spawn path below)
-    (void)kids;
 
     if (rank == -2) {
         // self-spawn path
