@@ -214,20 +214,19 @@ int factor_panel(Ctx &c, RankState &r, int n, std::vector<int> &ipiv_out) {
         const int nb = std::min(NB, nsteps - jb);
         const int m = n - jb;  // rows of the sub-panel
         if (launch_panel_factor(r.panel + i64(jb) * v + jb, v, m, nb, r.sync,
-                                r.d_ipiv + jb, c.epoch, c.stream)) {
+                                r.d_ipiv + jb, c.epoch, r.d_swap,
+                                r.d_swap + 64, c.stream)) {
             c.err = "panel grid not resident";
             return CONFLUX_LU_EINTERNAL;
         }
         c.epoch += nb;
-        // apply the sub-panel's swaps to the rest of the panel width:
-        // build the realized row permutation once, then two parallel passes
-        // over the non-sub-panel columns (vs nswap serialized round trips)
-        if (v > nb) {
-            launch_swap_map(r.d_ipiv + jb, nb, jb, r.d_swap, r.d_swap + 64,
-                            c.stream);
-            launch_rowperm_skip(r.panel, v, r.d_swap, r.d_swap + 64, 2 * nb,
-                                jb, nb, v - nb, r.rowtmp, c.stream);
-        }
+        // apply the sub-panel's swaps to the rest of the panel width: the
+        // factor kernel composed the realized row permutation (sub-panel-
+        // relative), applied here in two parallel passes over the
+        // non-sub-panel columns
+        if (v > nb)
+            launch_rowperm_skip(r.panel, v, r.d_swap, r.d_swap + 64, jb,
+                                2 * nb, jb, nb, v - nb, r.rowtmp, c.stream);
         if (jb + nb < v && m > nb) {
             // U block: rows jb..jb+nb of cols jb+nb..v
             launch_trsm_left_lower_unit32(r.panel + i64(jb) * v + jb, v,

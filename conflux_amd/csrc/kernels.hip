@@ -130,85 +130,31 @@ __global__ void k_row_scatter(const double *__restrict__ src, int64_t lds,
 // engine always stages through separate buffers (3-phase like the reference).
 
 
-// Build the row-permutation the dlaswp swap sequence realizes: after
-// applying swaps (i0+s <-> i0+piv[s]) in order, row dst_idx[r] holds the
-// original content of row src_idx[r].  2*nb entries, identity-padded.
-// Single-thread kernel (<= 64*64 scalar ops) so laswp becomes two parallel
-// passes instead of nswap serialized global round trips per column.
-__global__ void k_swap_map(const int *__restrict__ piv, int nb, int i0,
-                           int *__restrict__ dst_idx,
-                           int *__restrict__ src_idx) {
-    // LDS state (a private array would spill to scratch — rule 20):
-    // rows [i0, i0+nb) are direct-indexed; rows below the sub-panel head go
-    // to a small overflow list (at most nb entries).
-    __shared__ int base_src[32];
-    __shared__ int ovf_pos[32], ovf_src[32];
-    if (threadIdx.x != 0 || blockIdx.x != 0) return;
-    for (int i = 0; i < nb; ++i) base_src[i] = i0 + i;
-    int novf = 0;
-    for (int s = 0; s < nb; ++s) {
-        const int p = i0 + piv[s];
-        int *b;
-        if (p < i0 + nb) {
-            b = &base_src[p - i0];
-        } else {
-            int j = 0;
-            for (; j < novf && ovf_pos[j] != p; ++j) {
-            }
-            if (j == novf) {
-                ovf_pos[j] = p;
-                ovf_src[j] = p;
-                ++novf;
-            }
-            b = &ovf_src[j];
-        }
-        const int t = base_src[s];
-        base_src[s] = *b;
-        *b = t;
-    }
-    for (int i = 0; i < 2 * nb; ++i) {
-        // pad by duplicating entry 0 (same dst, same src: benign identical
-        // concurrent writes) so the launch geometry can be static
-        int d, sv;
-        if (i < nb) {
-            d = i0 + i;
-            sv = base_src[i];
-        } else if (i - nb < novf) {
-            d = ovf_pos[i - nb];
-            sv = ovf_src[i - nb];
-        } else {
-            d = i0;
-            sv = base_src[0];
-        }
-        dst_idx[i] = d;
-        src_idx[i] = sv;
-    }
-}
 
 // gather/scatter with a skipped column range [skip0, skip0+skipn): the
 // sub-panel columns were already swapped inside k_panel_factor.
 __global__ void k_rowperm_gather_skip(const double *__restrict__ src,
                                       int64_t lds, double *__restrict__ tmp,
                                       const int *__restrict__ src_idx,
-                                      int n_rows, int64_t skip0, int64_t skipn,
-                                      int64_t tot_cols) {
+                                      int row_base, int n_rows, int64_t skip0,
+                                      int64_t skipn, int64_t tot_cols) {
     const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= (int64_t)n_rows * tot_cols) return;
     const int64_t r = i / tot_cols, cc = i % tot_cols;
     const int64_t c = (cc < skip0) ? cc : cc + skipn;
-    tmp[r * tot_cols + cc] = src[(int64_t)src_idx[r] * lds + c];
+    tmp[r * tot_cols + cc] = src[(int64_t)(row_base + src_idx[r]) * lds + c];
 }
 
 __global__ void k_rowperm_scatter_skip(const double *__restrict__ tmp,
                                        double *__restrict__ dst, int64_t ldd,
                                        const int *__restrict__ dst_idx,
-                                       int n_rows, int64_t skip0,
+                                       int row_base, int n_rows, int64_t skip0,
                                        int64_t skipn, int64_t tot_cols) {
     const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= (int64_t)n_rows * tot_cols) return;
     const int64_t r = i / tot_cols, cc = i % tot_cols;
     const int64_t c = (cc < skip0) ? cc : cc + skipn;
-    dst[(int64_t)dst_idx[r] * ldd + c] = tmp[r * tot_cols + cc];
+    dst[(int64_t)(dst_idx[r] + row_base) * ldd + c] = tmp[r * tot_cols + cc];
 }
 
 // dst row dst_idx[i] <- src row src_idx[i]; row sets must be disjoint
@@ -255,12 +201,17 @@ __global__ void k_row_move(const double *__restrict__ src, int64_t lds,
 // matrix memory, no fences on the bulk (G16 R1: sc1 payload -> drain ->
 // flag; consumers poll relaxed + read sc1).
 // --------------------------------------------------------------------------
+// All per-column slabs are DOUBLE-BUFFERED by column parity: a fast block
+// may publish column c+1 while a slow one still reads column c (it cannot
+// reach c+2's publish before every block published c+1, so two slots
+// suffice).  Single-buffered slabs raced (old flag + new payload is
+// observable: no ordering between one block's reads and another's writes).
 struct PanelSync2 {
-    unsigned long long key_abs[CONFLUX_PANEL_MAX_BLOCKS];
-    unsigned long long key_flag[CONFLUX_PANEL_MAX_BLOCKS];  // (epoch<<32)|row
-    double cand_row[CONFLUX_PANEL_MAX_BLOCKS][PANEL_NB];
-    double diag_row[PANEL_NB];
-    unsigned int diag_flag;
+    unsigned long long key_abs[2][CONFLUX_PANEL_MAX_BLOCKS];
+    unsigned long long key_flag[2][CONFLUX_PANEL_MAX_BLOCKS];  // (epoch<<32)|row
+    double cand_row[2][CONFLUX_PANEL_MAX_BLOCKS][PANEL_NB];
+    double diag_row[2][PANEL_NB];
+    unsigned int diag_flag[2];
     unsigned int err;
 };
 
@@ -339,7 +290,8 @@ __global__ __launch_bounds__(256) void k_trsm_left_lower_unit32(
 __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     double *__restrict__ panel, int64_t ldp, int m, int nb,
     PanelSync2 *__restrict__ sync, int *__restrict__ ipiv,
-    unsigned int epoch0, int nblocks) {
+    unsigned int epoch0, int nblocks, int *__restrict__ swap_dst,
+    int *__restrict__ swap_src) {
     const int tid = threadIdx.x, bid = blockIdx.x;
     const int r0 = bid * PANEL_RPB + tid;
     const int r1 = r0 + PANEL_TPB;
@@ -350,10 +302,17 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     __shared__ int red_row[PANEL_TPB];
     __shared__ unsigned int sh_info[2];
     __shared__ int sh_pub[4];  // winner (tid, q), diag owner (tid, q)
+    // block 0 lane 0 composes the dlaswp row-permutation incrementally as
+    // pivots are decided (replaces the separate k_swap_map launch):
+    __shared__ int sm_base[PANEL_NB];
+    __shared__ int sm_opos[PANEL_NB], sm_osrc[PANEL_NB];
+    __shared__ int sm_novf;
+    if (bid == 0 && tid == 0) {
+        for (int i = 0; i < nb; ++i) sm_base[i] = i;
+        sm_novf = 0;
+    }
     const auto srsrc = __builtin_amdgcn_make_buffer_rsrc(
         (void *)sync, (short)0, (int)sizeof(PanelSync2), 0x00020000);
-    const int cand_off = (int)offsetof(PanelSync2, cand_row) + bid * PANEL_NB * 8;
-    const int diag_off = (int)offsetof(PanelSync2, diag_row);
 
     for (int q = 0; q < 2; ++q) {
         const int r = q ? r1 : r0;
@@ -363,6 +322,11 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
 
     for (int c = 0; c < nb; ++c) {
         const unsigned int epoch = epoch0 + (unsigned)c;
+        const int par = (int)(epoch & 1u);  // slab parity slot
+        const int cand_off = (int)offsetof(PanelSync2, cand_row) +
+                             (par * CONFLUX_PANEL_MAX_BLOCKS + bid) * PANEL_NB * 8;
+        const int diag_off =
+            (int)offsetof(PanelSync2, diag_row) + par * PANEL_NB * 8;
         // ---- local candidate: first-max over own rows >= c --------------
         double amax = -1.0;
         int arow = m;
@@ -429,15 +393,15 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         if (tid == 0) {
             union { double d; unsigned long long u; } a;
             a.d = wa;
-            st_rlx_u64(&sync->key_abs[bid], a.u);
+            st_rlx_u64(&sync->key_abs[par][bid], a.u);
             drain_stores();
             // flag granule carries the row (R2: data IS the tag)
-            st_rlx_u64(&sync->key_flag[bid],
+            st_rlx_u64(&sync->key_flag[par][bid],
                        ((unsigned long long)epoch << 32) | (unsigned)wrow);
             // the block owning row c raises diag_flag (its storing wave
             // drained before the barrier above)
             if (c >= bid * PANEL_RPB && c < (bid + 1) * PANEL_RPB)
-                st_rlx_u32(&sync->diag_flag, epoch);
+                st_rlx_u32(&sync->diag_flag[par], epoch);
         }
 
         // ---- EVERY block reduces the global winner itself (redundant,
@@ -450,7 +414,8 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
             for (int b = tid; b < nblocks; b += 64) {
                 unsigned long long g;
                 unsigned spins = 0;
-                while (((g = ld_rlx_u64(&sync->key_flag[b])) >> 32) != epoch) {
+                while (((g = ld_rlx_u64(&sync->key_flag[par][b])) >> 32) !=
+                       epoch) {
                     __builtin_amdgcn_s_sleep(1);
                     if (++spins > 800000000u) {
                         st_rlx_u32(&sync->err, 1u + (unsigned)c);
@@ -458,7 +423,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                     }
                 }
                 union { double d; unsigned long long u; } a;
-                a.u = ld_rlx_u64(&sync->key_abs[b]);
+                a.u = ld_rlx_u64(&sync->key_abs[par][b]);
                 const int rr = (int)(g & 0xffffffffu);
                 if (a.d > a_d || (a.d == a_d && rr < a_row)) {
                     a_d = a.d;
@@ -477,12 +442,32 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                 }
             }
             if (tid == 0) {
-                if (bid == 0) ipiv[c] = a_row;
+                if (bid == 0) {
+                    ipiv[c] = a_row;
+                    // compose the swap (c <-> a_row) into the row map
+                    int *b;
+                    if (a_row < nb) {
+                        b = &sm_base[a_row];
+                    } else {
+                        int j = 0;
+                        for (; j < sm_novf && sm_opos[j] != a_row; ++j) {
+                        }
+                        if (j == sm_novf) {
+                            sm_opos[j] = a_row;
+                            sm_osrc[j] = a_row;
+                            ++sm_novf;
+                        }
+                        b = &sm_osrc[j];
+                    }
+                    const int t = sm_base[c];
+                    sm_base[c] = *b;
+                    *b = t;
+                }
                 sh_info[0] = (unsigned)a_row;
                 sh_info[1] = (unsigned)a_win;
                 // wait the diagonal row publication too
                 unsigned spins = 0;
-                while (ld_rlx_u32(&sync->diag_flag) != epoch) {
+                while (ld_rlx_u32(&sync->diag_flag[par]) != epoch) {
                     __builtin_amdgcn_s_sleep(1);
                     if (++spins > 800000000u) {
                         st_rlx_u32(&sync->err, 1000000u + (unsigned)c);
@@ -498,7 +483,8 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
             F64x2Bits x;
             x.v = __builtin_amdgcn_raw_buffer_load_b128(
                 srsrc,
-                (int)offsetof(PanelSync2, cand_row) + win * PANEL_NB * 8 +
+                (int)offsetof(PanelSync2, cand_row) +
+                    (par * CONFLUX_PANEL_MAX_BLOCKS + win) * PANEL_NB * 8 +
                     16 * tid,
                 0, /*sc1*/ 16);
             piv_lds[2 * tid] = x.d[0];
@@ -543,6 +529,29 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         if (r >= m) continue;
         for (int cc = 0; cc < nb; ++cc)
             panel[(int64_t)r * ldp + cc] = rows[q][tid][cc];
+    }
+    // block 0: emit the composed row-permutation map (identity-padded by
+    // duplicating entry 0 — benign identical concurrent writes downstream).
+    // sm_* visibility: lane 0's last update precedes the column loop's
+    // closing __syncthreads, so no extra barrier (and none would be legal
+    // in this divergent tail).
+    if (bid == 0 && tid < 2 * PANEL_NB && swap_dst) {
+        const int i = tid;
+        int d, sv;
+        if (i < nb) {
+            d = i;
+            sv = sm_base[i];
+        } else if (i - nb < sm_novf) {
+            d = sm_opos[i - nb];
+            sv = sm_osrc[i - nb];
+        } else {
+            d = 0;
+            sv = sm_base[0];
+        }
+        if (i < 2 * nb) {
+            swap_dst[i] = d;
+            swap_src[i] = sv;
+        }
     }
 }
 
@@ -1259,34 +1268,30 @@ void launch_row_scatter(const double *src, int64_t lds, double *dst,
 
 
 
-void launch_swap_map(const int *piv, int nb, int i0, int *dst_idx,
-                     int *src_idx, hipStream_t s) {
-    hipLaunchKernelGGL(k_swap_map, dim3(1), dim3(64), 0, s, piv, nb, i0,
-                       dst_idx, src_idx);
-}
 
 void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
-                         const int *src_idx, int n_rows, int64_t skip0,
-                         int64_t skipn, int64_t tot_cols, double *tmp,
-                         hipStream_t s) {
+                         const int *src_idx, int row_base, int n_rows,
+                         int64_t skip0, int64_t skipn, int64_t tot_cols,
+                         double *tmp, hipStream_t s) {
     if (n_rows <= 0 || tot_cols <= 0) return;
     const int64_t n = (int64_t)n_rows * tot_cols;
     hipLaunchKernelGGL(k_rowperm_gather_skip, dim3(cdiv64(n, 256)), dim3(256),
-                       0, s, mat, ld, tmp, src_idx, n_rows, skip0, skipn,
-                       tot_cols);
+                       0, s, mat, ld, tmp, src_idx, row_base, n_rows, skip0,
+                       skipn, tot_cols);
     hipLaunchKernelGGL(k_rowperm_scatter_skip, dim3(cdiv64(n, 256)), dim3(256),
-                       0, s, tmp, mat, ld, dst_idx, n_rows, skip0, skipn,
-                       tot_cols);
+                       0, s, tmp, mat, ld, dst_idx, row_base, n_rows, skip0,
+                       skipn, tot_cols);
 }
 
 int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
-                        int *ipiv, unsigned int epoch0, hipStream_t s) {
+                        int *ipiv, unsigned int epoch0, int *swap_dst,
+                        int *swap_src, hipStream_t s) {
     int nblocks = (int)cdiv64(m, PANEL_RPB);
     if (nblocks < 1) nblocks = 1;
     if (nblocks > CONFLUX_PANEL_MAX_BLOCKS) return -1;  // not resident: refuse
     hipLaunchKernelGGL(k_panel_factor, dim3(nblocks), dim3(PANEL_TPB), 0, s,
                        panel, ldp, m, nb, (PanelSync2 *)sync, ipiv, epoch0,
-                       nblocks);
+                       nblocks, swap_dst, swap_src);
     return 0;
 }
 

@@ -21,14 +21,13 @@ void launch_row_gather(const double *src, int64_t lds, double *dst,
 void launch_row_scatter(const double *src, int64_t lds, double *dst,
                         int64_t ldd, const int *idx, int n_rows, int64_t cols,
                         hipStream_t s);
-void launch_swap_map(const int *piv, int nb, int i0, int *dst_idx,
-                     int *src_idx, hipStream_t s);
 void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
-                         const int *src_idx, int n_rows, int64_t skip0,
-                         int64_t skipn, int64_t tot_cols, double *tmp,
-                         hipStream_t s);
+                         const int *src_idx, int row_base, int n_rows,
+                         int64_t skip0, int64_t skipn, int64_t tot_cols,
+                         double *tmp, hipStream_t s);
 int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
-                        int *ipiv, unsigned int epoch0, hipStream_t s);
+                        int *ipiv, unsigned int epoch0, int *swap_dst,
+                        int *swap_src, hipStream_t s);
 int conflux_panel_sync_bytes();
 int conflux_panel_nb();
 void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
