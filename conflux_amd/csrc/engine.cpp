@@ -112,6 +112,8 @@ struct Ctx {
     hipStream_t stream{};
     hipStream_t panel_stream{};  // lookahead stream (distributed mode)
     hipEvent_t ev_pc{};          // panel-columns-updated event
+    hipEvent_t ev_t3{};          // step-3 complete (gates the step-5 TRSM)
+    hipEvent_t ev_t5{};          // step-5 TRSM complete (gates the C9 spread)
     std::vector<RankState> rs;   // size P (sim) or 1 (distributed)
     std::vector<int> pivotInds;  // M, global pivot ids (all ranks identical)
     unsigned epoch = 1;
@@ -857,6 +859,38 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         }
     }
 
+    // ---- steps 4 & 5 compute: the two panel TRSMs are independent after
+    // step 3 — run the step-5 TRSM on the second stream concurrently with
+    // the step-4 TRSM (the spreads below stay ordered on the main stream,
+    // C8 overlapping the step-5 solve like the reference's Iscatterv window,
+    // conflux_opt.hpp:1424-1615).
+    const bool split_trsm = !c.sim && c.panel_stream;
+    if (split_trsm) {
+        HIPCHK(hipEventRecord(c.ev_t3, c.stream));
+        HIPCHK(hipStreamWaitEvent(c.panel_stream, c.ev_t3, 0));
+        hipStream_t saved = c.stream;
+        c.stream = c.panel_stream;
+        for (auto &r : c.rs) {
+            if (r.pi != krow || r.pk != 0) continue;
+            if (trsm_left_lower(c, r, r.A01, wA01, wA01))
+                return CONFLUX_LU_EINTERNAL;
+            if (c.store_factors) {
+                const int ltik = k / Px;
+                const int64_t ustart =
+                    i64(v) * (r.pj < kcol ? k / Py + 1 : k / Py);
+                if (Nl - ustart > 0)
+                    launch_copy2d(r.A01 + (ustart - loff), wA01,
+                                  r.Fres + i64(ltik) * v * Nl + ustart, Nl, v,
+                                  Nl - ustart, c.stream);
+                if (r.pj == kcol)
+                    launch_copy2d(r.A00, v, r.Fres + i64(ltik) * v * Nl + loff,
+                                  Nl, v, v, c.stream);
+            }
+        }
+        HIPCHK(hipEventRecord(c.ev_t5, c.stream));
+        c.stream = saved;
+    }
+
     // ---- step 4: A10 <- A10 U^-1, slab-split, spread (C8) ------------------
     for (auto &r : c.rs) {
         if (r.pj != kcol || r.pk != 0) continue;
@@ -923,27 +957,33 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         }
     }
 
-    // ---- step 5: A01 <- L^-1 A01, spread (C9) ------------------------------
-    for (auto &r : c.rs) {
-        if (r.pi != krow || r.pk != 0) continue;
-        if (trsm_left_lower(c, r, r.A01, wA01, wA01))
-            return CONFLUX_LU_EINTERNAL;
-        if (c.store_factors) {
-            const int ltik = k / Px;
-            // U region starts at this rank's first local column tile with
-            // global tile >= k; columns left of it carry stale
-            // already-factored data the reference never reads
-            // (cf. oracle lu_oracle.py step-5 `gcs >= off` mask)
-            const int64_t ustart =
-                i64(v) * (r.pj < kcol ? k / Py + 1 : k / Py);
-            if (Nl - ustart > 0)
-                launch_copy2d(r.A01 + (ustart - loff), wA01,
-                              r.Fres + i64(ltik) * v * Nl + ustart, Nl, v,
-                              Nl - ustart, c.stream);
-            if (r.pj == kcol)  // diagonal tile: packed LU from A00
-                launch_copy2d(r.A00, v, r.Fres + i64(ltik) * v * Nl + loff, Nl,
-                              v, v, c.stream);
+    // ---- step 5: A01 <- L^-1 A01 (already issued on the second stream
+    // when split_trsm; the C9 spread below waits for it), spread (C9) -------
+    if (!split_trsm) {
+        for (auto &r : c.rs) {
+            if (r.pi != krow || r.pk != 0) continue;
+            if (trsm_left_lower(c, r, r.A01, wA01, wA01))
+                return CONFLUX_LU_EINTERNAL;
+            if (c.store_factors) {
+                const int ltik = k / Px;
+                // U region starts at this rank's first local column tile
+                // with global tile >= k; columns left of it carry stale
+                // already-factored data the reference never reads
+                // (cf. oracle lu_oracle.py step-5 `gcs >= off` mask)
+                const int64_t ustart =
+                    i64(v) * (r.pj < kcol ? k / Py + 1 : k / Py);
+                if (Nl - ustart > 0)
+                    launch_copy2d(r.A01 + (ustart - loff), wA01,
+                                  r.Fres + i64(ltik) * v * Nl + ustart, Nl, v,
+                                  Nl - ustart, c.stream);
+                if (r.pj == kcol)  // diagonal tile: packed LU from A00
+                    launch_copy2d(r.A00, v,
+                                  r.Fres + i64(ltik) * v * Nl + loff, Nl, v,
+                                  v, c.stream);
+            }
         }
+    } else {
+        HIPCHK(hipStreamWaitEvent(c.stream, c.ev_t5, 0));
     }
     if (Pz == 1 && Px == 1) {
         RankState &r = c.rs[0];
@@ -1443,7 +1483,9 @@ int conflux_lu_create(int N, int v, int Px, int Py, int Pz, int rank,
         const int pi = rank / (Py * Pz), pj = (rank / Pz) % Py, pk = rank % Pz;
         if (alloc_rank(*c, c->rs[0], pi, pj, pk)) { delete c; return CONFLUX_LU_EHIP; }
         if (hipStreamCreate(&c->panel_stream) != hipSuccess ||
-            hipEventCreate(&c->ev_pc) != hipSuccess) {
+            hipEventCreate(&c->ev_pc) != hipSuccess ||
+            hipEventCreate(&c->ev_t3) != hipSuccess ||
+            hipEventCreate(&c->ev_t5) != hipSuccess) {
             delete c;
             return CONFLUX_LU_EHIP;
         }
@@ -1704,6 +1746,8 @@ int conflux_lu_destroy(conflux_lu_ctx *c) {
     if (c->have_comm) (void)ncclCommDestroy(c->comm);
     if (c->panel_stream) (void)hipStreamDestroy(c->panel_stream);
     if (c->ev_pc) (void)hipEventDestroy(c->ev_pc);
+    if (c->ev_t3) (void)hipEventDestroy(c->ev_t3);
+    if (c->ev_t5) (void)hipEventDestroy(c->ev_t5);
     (void)hipStreamDestroy(c->stream);
     delete c;
     return CONFLUX_LU_OK;
