@@ -432,6 +432,7 @@ int factor_loop(Ctx &c, double *elapsed_ms) {
         HIPCHK(hipStreamSynchronize(c.stream));
     }
     const auto t2 = std::chrono::high_resolution_clock::now();
+    HIPCHK(hipGetLastError());  // surface any failed launch loudly
     if (elapsed_ms)
         *elapsed_ms =
             std::chrono::duration<double, std::milli>(t2 - t1).count();
@@ -1401,6 +1402,7 @@ int chol_loop(Ctx &c, double *elapsed_ms) {
         HIPCHK(hipStreamSynchronize(c.stream));
     }
     const auto t2 = std::chrono::high_resolution_clock::now();
+    HIPCHK(hipGetLastError());  // surface any failed launch loudly
     if (elapsed_ms)
         *elapsed_ms =
             std::chrono::duration<double, std::milli>(t2 - t1).count();
