@@ -65,7 +65,7 @@ def cpu_baseline():
         out = subprocess.run(
             ["/opt/conda/bin/mpiexec", "-n", str(n_ranks), ref, str(N),
              str(v), grid[0], grid[1], grid[2], "-", "/tmp/confluxref_bench",
-             "1"],
+             "2"],
             env=env, capture_output=True, text=True, timeout=900)
         ms = None
         for line in out.stdout.splitlines():
@@ -122,6 +122,18 @@ def main():
         objs = [Engine.make_uid()] if rank == 0 else [None]
         dist.broadcast_object_list(objs, src=0)
         uid = objs[0]
+
+    if os.environ.get("CONFLUX_BENCH_DRYRUN"):
+        # bootstrap-only validation (CI on GPU-less hosts): arg parsing,
+        # gloo init, uid broadcast — everything up to engine creation
+        if rank == 0:
+            print(json.dumps({"dryrun": True, "n_gpus": n_gpus, "N": N,
+                              "v": v, "grid": f"{Px}x{Py}x{Pz}",
+                              "uid_ok": uid is None or len(uid) == 128}))
+        if dist:
+            dist.barrier()
+            dist.destroy_process_group()
+        return
 
     eng = Engine(N, v, Px, Py, Pz, rank=(0 if world == 1 else rank),
                  world=world, uid=uid)
