@@ -1,0 +1,29 @@
+"""C-ABI argument guards (no GPU needed: rejected before any HIP call)."""
+import ctypes
+
+import pytest
+
+
+@pytest.fixture(scope="module")
+def lib():
+    import conflux_amd
+    try:
+        return conflux_amd.lib()
+    except conflux_amd.ConfluxLuError:
+        pytest.skip("HIP engine not built")
+
+
+@pytest.mark.parametrize("args", [
+    (256, 32, 2, 4, 1),   # Px != Py
+    (256, 32, 3, 3, 1),   # non-power-of-two Px
+    (256, 33, 1, 1, 2),   # v % Pz != 0
+    (250, 32, 1, 1, 1),   # N not a multiple of v*Px
+    (64, 64, 1, 1, 1),    # Ml < 2v (reference buffer-sizing envelope)
+    (0, 32, 1, 1, 1),     # degenerate
+])
+def test_create_rejects_unsupported_grids(lib, args):
+    N, v, Px, Py, Pz = args
+    h = ctypes.c_void_p()
+    rc = lib.conflux_lu_create(N, v, Px, Py, Pz, -1, Px * Py * Pz, None,
+                               ctypes.byref(h))
+    assert rc == -1  # CONFLUX_LU_EARG
