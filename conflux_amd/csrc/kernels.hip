@@ -301,7 +301,6 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     __shared__ double red_abs[PANEL_TPB];
     __shared__ int red_row[PANEL_TPB];
     __shared__ unsigned int sh_info[2];
-    __shared__ int sh_pub[4];  // winner (tid, q), diag owner (tid, q)
     // block 0 lane 0 composes the dlaswp row-permutation incrementally as
     // pivots are decided (replaces the separate k_swap_map launch):
     __shared__ int sm_base[PANEL_NB];
@@ -356,35 +355,32 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
             const int orr = red_row[wv];
             if (oa > wa || (oa == wa && orr < wrow)) { wa = oa; wrow = orr; }
         }
-        // ---- cooperative publication: owners of the winner row and the
-        // diagonal row mark their (tid, q); lanes 0-15 stream the winner
-        // row, lanes 16-31 the diag row, as 16-byte sc1 buffer stores; all
-        // publishing lanes are wave 0, so one wave drain covers them.
-        if (wrow < m && (wrow == r0 || wrow == r1)) {
-            sh_pub[0] = tid;
-            sh_pub[1] = (wrow == r1);
-        } else if (wrow >= m && tid == 0) {
-            sh_pub[0] = 0;  // empty block: junk row; its key_abs = -1 loses
-            sh_pub[1] = 0;
-        }
-        if (c == r0 || c == r1) {
-            sh_pub[2] = tid;
-            sh_pub[3] = (c == r1);
-        }
-        __syncthreads();
+        // ---- cooperative publication: lanes 0-15 stream the winner row,
+        // lanes 16-31 the diag row, as 16-byte sc1 buffer stores.  The
+        // owning (thread, q) of any row r is computable from r directly
+        // (r = bid*RPB + tid + q*TPB), so no extra LDS handoff or barrier:
+        // the publishing lanes read the owner's LDS slot (fresh — last
+        // written before the column loop's closing barrier).  All
+        // publishing lanes are wave 0; one wave drain covers them.
         if (tid < 16) {
+            const int lrow = (wrow < m) ? wrow - bid * PANEL_RPB : 0;
+            const int wq = lrow >= PANEL_TPB;
+            const int wtid = lrow - wq * PANEL_TPB;
             F64x2Bits x;
-            x.d[0] = rows[sh_pub[1]][sh_pub[0]][2 * tid];
-            x.d[1] = rows[sh_pub[1]][sh_pub[0]][2 * tid + 1];
+            x.d[0] = rows[wq][wtid][2 * tid];
+            x.d[1] = rows[wq][wtid][2 * tid + 1];
             if (2 * tid < nb)
                 __builtin_amdgcn_raw_buffer_store_b128(
                     x.v, srsrc, cand_off + 16 * tid, 0, /*sc1*/ 16);
         } else if (tid < 32 && c >= bid * PANEL_RPB &&
                    c < (bid + 1) * PANEL_RPB) {
             const int l = tid - 16;
+            const int lrow = c - bid * PANEL_RPB;
+            const int dq = lrow >= PANEL_TPB;
+            const int dtid = lrow - dq * PANEL_TPB;
             F64x2Bits x;
-            x.d[0] = rows[sh_pub[3]][sh_pub[2]][2 * l];
-            x.d[1] = rows[sh_pub[3]][sh_pub[2]][2 * l + 1];
+            x.d[0] = rows[dq][dtid][2 * l];
+            x.d[1] = rows[dq][dtid][2 * l + 1];
             if (2 * l < nb)
                 __builtin_amdgcn_raw_buffer_store_b128(
                     x.v, srsrc, diag_off + 16 * l, 0, /*sc1*/ 16);
@@ -416,7 +412,6 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                 unsigned spins = 0;
                 while (((g = ld_rlx_u64(&sync->key_flag[par][b])) >> 32) !=
                        epoch) {
-                    __builtin_amdgcn_s_sleep(1);
                     if (++spins > 800000000u) {
                         st_rlx_u32(&sync->err, 1u + (unsigned)c);
                         break;
