@@ -1035,46 +1035,63 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
     }
 
     // ---- step 6: trailing update (the flop carrier), split for lookahead --
-    auto gemm_piece = [&](RankState &r, int64_t col0, int64_t ncols) -> int {
+    auto gemm_piece = [&](RankState &r, int64_t col0, int64_t ncols,
+                          int cap) -> int {
         if (r.nact <= 0 || ncols <= 0) return 0;
         const double fl = 2.0 * r.nact * (double)ncols * c.nlayr;
         size_t slot;
         if (ev_begin(c, 0, fl, &slot)) return CONFLUX_LU_EHIP;
         launch_dgemm_f64(r.A10Rcv, c.nlayr, r.A01Rcv + (col0 - loff), Nl,
                          r.A11 + i64(r.fnp) * Nl + col0, Nl, r.nact, ncols,
-                         c.nlayr, c.stream);
+                         c.nlayr, c.stream, cap);
         return ev_end(c, slot) ? CONFLUX_LU_EHIP : 0;
     };
     const bool look = (k + 1 < c.Nt);
     const int ncol = (k + 1) % Py;
     const int64_t lnext = i64(v) * ((k + 1) / Py);
+    const char *lk = getenv("CONFLUX_LOOKAHEAD");
+    const bool async_look = look && !c.sim && c.panel_stream &&
+                            (lk ? atoi(lk) != 0 : true);
+    // While the panel of step k+1 runs concurrently, cap the trailing
+    // update's grid so whole CUs stay free for it: a panel block needs
+    // 135 KB LDS (a full CU), so an uncapped GEMM flood starves the panel
+    // until the queue drains (measured 256 ms vs 251 ms sequential at
+    // N=16384 before this cap).  Default leaves 40 of 256 CUs free.
+    int gcap = 0;
+    if (async_look) {
+        static int env_cap = -1;
+        if (env_cap < 0) {
+            const char *e = getenv("CONFLUX_GEMM_CAP");
+            env_cap = e ? atoi(e) : 432;
+        }
+        gcap = env_cap;
+    }
     // (a) the columns step k+1's panel needs, first
     if (look)
         for (auto &r : c.rs)
             if (r.pj == ncol)
-                if (gemm_piece(r, lnext, v)) return CONFLUX_LU_EHIP;
+                if (gemm_piece(r, lnext, v, 0)) return CONFLUX_LU_EHIP;
     if (look && !c.sim && c.panel_stream) {
         HIPCHK(hipEventRecord(c.ev_pc, c.stream));
     }
-    // (b) the rest of the trailing update
+    // (b) the rest of the trailing update — capped while the panel runs.
+    // It must be ENQUEUED before phase01: phase01 ends in a host sync (the
+    // pivot D2H), so anything enqueued after it cannot overlap the panel.
     for (auto &r : c.rs) {
         if (look && r.pj == ncol) {
-            if (gemm_piece(r, loff, lnext - loff)) return CONFLUX_LU_EHIP;
-            if (gemm_piece(r, lnext + v, Nl - (lnext + v))) return CONFLUX_LU_EHIP;
+            if (gemm_piece(r, loff, lnext - loff, gcap)) return CONFLUX_LU_EHIP;
+            if (gemm_piece(r, lnext + v, Nl - (lnext + v), gcap))
+                return CONFLUX_LU_EHIP;
         } else {
-            if (gemm_piece(r, loff, wA01)) return CONFLUX_LU_EHIP;
+            if (gemm_piece(r, loff, wA01, gcap)) return CONFLUX_LU_EHIP;
         }
     }
-    // (c) lookahead: step k+1's panel chain, overlapped with (b).
-    // The overlap engages only for world > 1 (it frees non-participant
-    // ranks); on a single GPU the latency-bound panel handshake slows more
-    // under GEMM load than the overlap saves (measured: 256 vs 251 ms at
-    // N=16384), so world == 1 keeps the sequential order on one stream.
+    // (c) lookahead: step k+1's panel chain on the panel stream, gated on
+    // ev_pc (= its columns updated), running concurrently with (b) on the
+    // CUs the cap left free.  Numerically identical: the column split does
+    // not reorder any K-sum.
     if (look) {
         hipStream_t saved = c.stream;
-        const char *lk = getenv("CONFLUX_LOOKAHEAD");
-        const bool async_look =
-            !c.sim && c.panel_stream && (lk ? atoi(lk) != 0 : c.world > 1);
         if (async_look) {
             HIPCHK(hipStreamWaitEvent(c.panel_stream, c.ev_pc, 0));
             c.stream = c.panel_stream;

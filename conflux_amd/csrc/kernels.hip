@@ -730,22 +730,26 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
     const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
     int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
     int ntm, int ntn) {
+    __shared__ double As[2][GEMM_BK][GEMM_BM + 1];   // transposed, padded
+    __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];   // double-buffered
+    const int nwg = ntm * ntn;
+    // persistent grid-stride over tiles: gridDim may be capped below nwg so
+    // the concurrent panel kernel (needs whole CUs: 135 KB LDS/block) can
+    // co-schedule during lookahead.  gridDim == nwg -> one iteration,
+    // identical to the plain launch.
+    for (int vwg = blockIdx.x; vwg < nwg; vwg += gridDim.x) {
     // bijective XCD swizzle (guide §5: q/r form)
-    int wg = blockIdx.x;
+    int wg = vwg;
     {
-        const int nwg = ntm * ntn;
         const int q = nwg >> 3, r = nwg & 7;
-        const int xcd = wg & 7, idx = wg >> 3;
+        const int xcd = vwg & 7, idx = vwg >> 3;
         // inverse of dispatch round-robin: give each XCD a contiguous chunk
         wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
-        if (nwg < 8) wg = blockIdx.x;   // tiny grids: identity
+        if (nwg < 8) wg = vwg;          // tiny grids: identity
     }
     const int tm = wg / ntn, tn = wg % ntn;
     const int row0 = tm * GEMM_BM;
     const int64_t col0 = (int64_t)tn * GEMM_BN;
-
-    __shared__ double As[2][GEMM_BK][GEMM_BM + 1];   // transposed, padded
-    __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];   // double-buffered
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -849,6 +853,10 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
                 }
             }
         }
+    }
+    // next tile re-stages buffer 0; the kt-loop's closing barrier already
+    // ordered every wave's last LDS read before it
+    if (vwg + gridDim.x < nwg) __syncthreads();
     }
 }
 
@@ -1318,7 +1326,7 @@ int g_dgemm_variant = -1;  // 0 = 4-wave, 1 = 8-wave; env CONFLUX_GEMM_VARIANT
 
 void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
                       int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
-                      int K, hipStream_t s) {
+                      int K, hipStream_t s, int maxwg) {
     if (M <= 0 || N <= 0 || K <= 0) return;
     if (g_dgemm_variant < 0) {
         const char *e = getenv("CONFLUX_GEMM_VARIANT");
@@ -1326,11 +1334,14 @@ void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
     }
     const int ntm = (int)cdiv64(M, GEMM_BM);
     const int ntn = (int)cdiv64(N, GEMM_BN);
+    int nwg = ntm * ntn;
+    // cap (persistent w8 kernel only): leave CUs free for a concurrent panel
+    if (maxwg > 0 && g_dgemm_variant == 1 && nwg > maxwg) nwg = maxwg;
     if (g_dgemm_variant == 2)
         hipLaunchKernelGGL(k_dgemm_f64_glds, dim3(ntm * ntn), dim3(512), 0, s,
                            A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
     else if (g_dgemm_variant == 1)
-        hipLaunchKernelGGL(k_dgemm_f64_w8, dim3(ntm * ntn), dim3(512), 0, s,
+        hipLaunchKernelGGL(k_dgemm_f64_w8, dim3(nwg), dim3(512), 0, s,
                            A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
     else
         hipLaunchKernelGGL(k_dgemm_f64, dim3(ntm * ntn), dim3(GEMM_TPB), 0, s,

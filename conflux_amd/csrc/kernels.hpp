@@ -45,7 +45,8 @@ void launch_init_matrix_spd(double *A, int Ml, int Nl, int v, int Px, int Py,
                             int64_t Nglob, hipStream_t s);
 void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
                       int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
-                      int K, hipStream_t s);
+                      int K, hipStream_t s,
+                      int maxwg = 0);
 void launch_pack_candidate(const double *A10, int64_t lda, const int *gri,
                            int f, int n_src, int n_out, int v, const int *idx,
                            double *cand, hipStream_t s);
