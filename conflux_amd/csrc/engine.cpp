@@ -1050,8 +1050,8 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
     const int ncol = (k + 1) % Py;
     const int64_t lnext = i64(v) * ((k + 1) / Py);
     const char *lk = getenv("CONFLUX_LOOKAHEAD");
-    const bool async_look = look && !c.sim && c.panel_stream &&
-                            (lk ? atoi(lk) != 0 : true);
+    bool async_look = look && !c.sim && c.panel_stream &&
+                      (lk ? atoi(lk) != 0 : true);
     // While the panel of step k+1 runs concurrently, cap the trailing
     // update's grid so whole CUs stay free for it: a panel block needs
     // 135 KB LDS (a full CU), so an uncapped GEMM flood starves the panel
@@ -1065,6 +1065,20 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
             env_cap = e ? atoi(e) : 432;
         }
         gcap = env_cap;
+        // Residency guard: the persistent capped GEMM holds its CUs for the
+        // whole update, so if step k+1's panel needs more co-resident
+        // blocks than the cap leaves free (e.g. tall panels at N=32768),
+        // the overlap would serialize anyway — keep the sequential order
+        // and the full-width GEMM instead.
+        int max_rows = 0;
+        for (auto &r : c.rs) max_rows = std::max(max_rows, r.nact);
+        const int nblocks = (max_rows + conflux_panel_rpb() - 1) /
+                            conflux_panel_rpb();
+        const int free_cus = 256 - (gcap + 1) / 2;
+        if (nblocks > free_cus) {
+            async_look = false;
+            gcap = 0;
+        }
     }
     // (a) the columns step k+1's panel needs, first
     if (look)

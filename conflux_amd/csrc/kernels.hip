@@ -1300,6 +1300,7 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
 
 int conflux_panel_sync_bytes() { return (int)sizeof(PanelSync2); }
 int conflux_panel_nb() { return PANEL_NB; }
+int conflux_panel_rpb() { return PANEL_RPB; }
 
 void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
                                    int64_t ldx, int nb, int64_t N,

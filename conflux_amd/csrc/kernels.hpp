@@ -30,6 +30,7 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
                         int *swap_src, hipStream_t s);
 int conflux_panel_sync_bytes();
 int conflux_panel_nb();
+int conflux_panel_rpb();
 void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
                                    int64_t ldx, int nb, int64_t N,
                                    hipStream_t s);
