@@ -267,6 +267,10 @@ int trsm_right_upper(Ctx &c, RankState &r, double *X, int64_t ldx, int M) {
     const int v = c.v, NB = conflux_panel_nb();
     size_t slot;
     if (ev_begin(c, 2, 0, &slot)) return CONFLUX_LU_EHIP;
+    if (v % 32 == 0) {  // fused single-launch MFMA solve
+        launch_trsm_right_mfma(r.A00, v, X, ldx, v, M, 0, c.stream);
+        return ev_end(c, slot);
+    }
     for (int jb = 0; jb < v; jb += NB) {
         const int nb = std::min(NB, v - jb);
         launch_trsm_right_upper32(r.A00 + i64(jb) * v + jb, v, X + jb, ldx, nb,
@@ -282,6 +286,10 @@ int trsm_left_lower(Ctx &c, RankState &r, double *X, int64_t ldx, int64_t N) {
     const int v = c.v, NB = conflux_panel_nb();
     size_t slot;
     if (ev_begin(c, 2, 0, &slot)) return CONFLUX_LU_EHIP;
+    if (v % 32 == 0) {  // fused single-launch MFMA solve
+        launch_trsm_left_mfma(r.A00, v, X, ldx, v, N, c.stream);
+        return ev_end(c, slot);
+    }
     for (int jb = 0; jb < v; jb += NB) {
         const int nb = std::min(NB, v - jb);
         launch_trsm_left_lower_unit32(r.A00 + i64(jb) * v + jb, v,
@@ -1164,6 +1172,10 @@ int trsm_right_lowT(Ctx &c, RankState &r, double *X, int64_t ldx, int M) {
     const int v = c.v, NB = conflux_panel_nb();
     size_t slot;
     if (ev_begin(c, 2, 0, &slot)) return CONFLUX_LU_EHIP;
+    if (v % 32 == 0) {  // fused single-launch MFMA solve (X * L^-T)
+        launch_trsm_right_mfma(r.A00, v, X, ldx, v, M, 1, c.stream);
+        return ev_end(c, slot);
+    }
     for (int jb = 0; jb < v; jb += NB) {
         const int nb = std::min(NB, v - jb);
         launch_trsm_right_upper32(r.A00 + i64(jb) * v + jb, v, X + jb, ldx, nb,

@@ -1318,6 +1318,269 @@ void launch_trsm_right_upper32(const double *U, int64_t ldu, double *X,
                        0, s, U, ldu, X, ldx, nb, M, trans);
 }
 
+// ---------------------------------------------------------------------------
+// Fused whole-panel TRSMs, MFMA edition: ONE launch applies the full v x v
+// triangle (v % 32 == 0).  Left-looking over 32-wide panels; the
+// rank-32 updates run on the matrix cores (v_mfma_f64_16x16x4_f64, the
+// lane maps verified by tools/mfma_probe), the 32x32 diagonal solves as
+// register triangles (the k_trsm_*32 pattern).  Each block owns a stripe
+// of X exclusively and stages panels through LDS, so the blocked chain's
+// 31 launches and its 16 HBM re-writes of X collapse into one kernel.
+// (A VALU version of this idea lost 3x to the blocked chain — the updates
+// belong on MFMA; see DESIGN.md ablations.)
+// ---------------------------------------------------------------------------
+
+// X (M x v, row-major) <- X * U^-1 (trans=0) or X * L^-T (trans=1).
+#define TRM_ROWS 128
+__global__ __launch_bounds__(256) void k_trsm_right_mfma(
+    const double *__restrict__ U, int64_t ldu, double *__restrict__ X,
+    int64_t ldx, int v, int64_t M, int trans) {
+    __shared__ double sXj[TRM_ROWS][33];     // current panel
+    __shared__ double sXp[2][TRM_ROWS][33];  // solved panels, double-buffered
+    __shared__ double sU[2][32][33];
+    const int tid = threadIdx.x;
+    const int lane = tid & 63, wave = tid >> 6;
+    const int frow = lane & 15, fk = lane >> 4;
+    const int64_t r0 = (int64_t)blockIdx.x * TRM_ROWS;
+    const int rows = (int)min((int64_t)TRM_ROWS, M - r0);
+    // register staging (w8-GEMM style): next panel's loads issue while MFMA
+    // consumes the current LDS buffer — one barrier per 32-panel, HBM
+    // latency hidden behind the update
+    double rx[16], ru[4];  // 128x32 X panel + 32x32 U block per iteration
+    auto load_u_regs = [&](int pb, int jb) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 256;
+            ru[i] = trans
+                        ? U[(int64_t)(jb + (e & 31)) * ldu + pb + (e >> 5)]
+                        : U[(int64_t)(pb + (e >> 5)) * ldu + jb + (e & 31)];
+        }
+    };
+    auto load_x_regs = [&](int pb) {
+#pragma unroll
+        for (int i = 0; i < 16; ++i) {
+            const int e = tid + i * 256;
+            rx[i] = (e >> 5) < rows
+                        ? X[(r0 + (e >> 5)) * ldx + pb + (e & 31)]
+                        : 0.0;
+        }
+    };
+    auto write_bufs = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < 16; ++i) {
+            const int e = tid + i * 256;
+            sXp[buf][e >> 5][e & 31] = rx[i];
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 256;
+            sU[buf][e >> 5][e & 31] = ru[i];
+        }
+    };
+    for (int jb = 0; jb < v; jb += 32) {
+        __syncthreads();
+        for (int i = tid; i < rows * 32; i += 256)
+            sXj[i >> 5][i & 31] = X[(r0 + (i >> 5)) * ldx + jb + (i & 31)];
+        int cur = 0;
+        if (jb > 0) {
+            load_x_regs(0);
+            load_u_regs(0, jb);
+            write_bufs(0);
+        }
+        __syncthreads();  // also covers the sXj stage
+        for (int pb = 0; pb < jb; pb += 32) {
+            if (pb + 32 < jb) {
+                load_x_regs(pb + 32);
+                load_u_regs(pb + 32, jb);
+            }
+            // sXj[rows x 32] -= sXp[rows x 32] * sU[32 x 32] on MFMA:
+            // wave w owns rows [32w, 32w+32): 2 row-tiles x 2 col-tiles
+            if (wave * 32 < rows) {
+#pragma unroll
+                for (int rt = 0; rt < 2; ++rt) {
+                    const int rbase = wave * 32 + rt * 16;
+                    f64x4 acc[2] = {f64x4{0, 0, 0, 0}, f64x4{0, 0, 0, 0}};
+#pragma unroll
+                    for (int kk = 0; kk < 8; ++kk) {
+                        const int k = kk * 4 + fk;
+                        const double a = sXp[cur][rbase + frow][k];
+#pragma unroll
+                        for (int ct = 0; ct < 2; ++ct)
+                            acc[ct] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                                a, sU[cur][k][ct * 16 + frow], acc[ct], 0, 0,
+                                0);
+                    }
+#pragma unroll
+                    for (int ct = 0; ct < 2; ++ct)
+#pragma unroll
+                        for (int q = 0; q < 4; ++q)
+                            sXj[rbase + 4 * q + fk][ct * 16 + frow] -=
+                                acc[ct][q];
+                }
+            }
+            if (pb + 32 < jb) write_bufs(cur ^ 1);
+            __syncthreads();
+            cur ^= 1;
+        }
+        // diagonal block into sU[0] for the solve
+        for (int i = tid; i < 32 * 32; i += 256)
+            sU[0][i >> 5][i & 31] =
+                trans ? U[(int64_t)(jb + (i & 31)) * ldu + jb + (i >> 5)]
+                      : U[(int64_t)(jb + (i >> 5)) * ldu + jb + (i & 31)];
+        __syncthreads();
+        if (tid < rows) {  // diagonal solve: thread = row, register triangle
+            double x[32];
+#pragma unroll
+            for (int c = 0; c < 32; ++c) x[c] = sXj[tid][c];
+#pragma unroll
+            for (int c = 0; c < 32; ++c) {
+                double acc = x[c];
+#pragma unroll
+                for (int i = 0; i < 32; ++i)
+                    if (i < c) acc -= x[i] * sU[0][i][c];
+                x[c] = acc / sU[0][c][c];
+            }
+#pragma unroll
+            for (int c = 0; c < 32; ++c) sXj[tid][c] = x[c];
+        }
+        __syncthreads();
+        for (int i = tid; i < rows * 32; i += 256)
+            X[(r0 + (i >> 5)) * ldx + jb + (i & 31)] = sXj[i >> 5][i & 31];
+    }
+}
+
+// X (v x N, row-major) <- L^-1 * X, L v x v unit lower.
+#define TRM_COLS 128
+__global__ __launch_bounds__(256) void k_trsm_left_mfma(
+    const double *__restrict__ L, int64_t ldl, double *__restrict__ X,
+    int64_t ldx, int v, int64_t N) {
+    __shared__ double sXj[32][TRM_COLS + 1];
+    __shared__ double sXp[2][32][TRM_COLS + 1];  // double-buffered
+    __shared__ double sL[2][32][33];
+    const int tid = threadIdx.x;
+    const int lane = tid & 63, wave = tid >> 6;
+    const int frow = lane & 15, fk = lane >> 4;
+    const int64_t c0 = (int64_t)blockIdx.x * TRM_COLS;
+    const int cols = (int)min((int64_t)TRM_COLS, N - c0);
+    double rx[16], rl[4];  // 32x128 X panel + 32x32 L block per iteration
+    auto load_l_regs = [&](int pb, int jb) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 256;
+            rl[i] = L[(int64_t)(jb + (e >> 5)) * ldl + pb + (e & 31)];
+        }
+    };
+    auto load_x_regs = [&](int pb) {
+#pragma unroll
+        for (int i = 0; i < 16; ++i) {
+            const int e = tid + i * 256;
+            rx[i] = (e & (TRM_COLS - 1)) < cols
+                        ? X[(int64_t)(pb + (e >> 7)) * ldx + c0 +
+                            (e & (TRM_COLS - 1))]
+                        : 0.0;
+        }
+    };
+    auto write_bufs = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < 16; ++i) {
+            const int e = tid + i * 256;
+            sXp[buf][e >> 7][e & (TRM_COLS - 1)] = rx[i];
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 256;
+            sL[buf][e >> 5][e & 31] = rl[i];
+        }
+    };
+    for (int jb = 0; jb < v; jb += 32) {
+        __syncthreads();
+        for (int i = tid; i < 32 * TRM_COLS; i += 256)
+            if ((i & (TRM_COLS - 1)) < cols)
+                sXj[i >> 7][i & (TRM_COLS - 1)] =
+                    X[(int64_t)(jb + (i >> 7)) * ldx + c0 +
+                      (i & (TRM_COLS - 1))];
+        int cur = 0;
+        if (jb > 0) {
+            load_x_regs(0);
+            load_l_regs(0, jb);
+            write_bufs(0);
+        }
+        __syncthreads();  // also covers the sXj stage
+        for (int pb = 0; pb < jb; pb += 32) {
+            if (pb + 32 < jb) {
+                load_x_regs(pb + 32);
+                load_l_regs(pb + 32, jb);
+            }
+            // sXj[32 x cols] -= sL[32 x 32] * sXp[32 x cols] on MFMA:
+            // wave w owns cols [32w, 32w+32): 2 row-tiles x 2 col-tiles
+            if (wave * 32 < cols) {
+#pragma unroll
+                for (int ct = 0; ct < 2; ++ct) {
+                    const int cbase = wave * 32 + ct * 16;
+                    f64x4 acc[2] = {f64x4{0, 0, 0, 0}, f64x4{0, 0, 0, 0}};
+#pragma unroll
+                    for (int kk = 0; kk < 8; ++kk) {
+                        const int k = kk * 4 + fk;
+                        const double b = sXp[cur][k][cbase + frow];
+#pragma unroll
+                        for (int rt = 0; rt < 2; ++rt)
+                            acc[rt] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                                sL[cur][rt * 16 + frow][k], b, acc[rt], 0, 0,
+                                0);
+                    }
+#pragma unroll
+                    for (int rt = 0; rt < 2; ++rt)
+#pragma unroll
+                        for (int q = 0; q < 4; ++q)
+                            sXj[rt * 16 + 4 * q + fk][cbase + frow] -=
+                                acc[rt][q];
+                }
+            }
+            if (pb + 32 < jb) write_bufs(cur ^ 1);
+            __syncthreads();
+            cur ^= 1;
+        }
+        for (int i = tid; i < 32 * 32; i += 256)
+            sL[0][i >> 5][i & 31] =
+                L[(int64_t)(jb + (i >> 5)) * ldl + jb + (i & 31)];
+        __syncthreads();
+        if (tid < cols) {  // diagonal solve: thread = column, unit lower
+            double x[32];
+#pragma unroll
+            for (int r = 0; r < 32; ++r) x[r] = sXj[r][tid];
+#pragma unroll
+            for (int k = 0; k < 32; ++k) {
+                const double xk = x[k];
+#pragma unroll
+                for (int r = 0; r < 32; ++r)
+                    if (r > k) x[r] -= sL[0][r][k] * xk;
+            }
+#pragma unroll
+            for (int r = 0; r < 32; ++r) sXj[r][tid] = x[r];
+        }
+        __syncthreads();
+        for (int i = tid; i < 32 * TRM_COLS; i += 256)
+            if ((i & (TRM_COLS - 1)) < cols)
+                X[(int64_t)(jb + (i >> 7)) * ldx + c0 + (i & (TRM_COLS - 1))] =
+                    sXj[i >> 7][i & (TRM_COLS - 1)];
+    }
+}
+
+void launch_trsm_right_mfma(const double *U, int64_t ldu, double *X,
+                            int64_t ldx, int v, int64_t M, int trans,
+                            hipStream_t s) {
+    if (M <= 0 || v <= 0) return;
+    hipLaunchKernelGGL(k_trsm_right_mfma, dim3(cdiv64(M, TRM_ROWS)),
+                       dim3(256), 0, s, U, ldu, X, ldx, v, M, trans);
+}
+
+void launch_trsm_left_mfma(const double *L, int64_t ldl, double *X,
+                           int64_t ldx, int v, int64_t N, hipStream_t s) {
+    if (N <= 0 || v <= 0) return;
+    hipLaunchKernelGGL(k_trsm_left_mfma, dim3(cdiv64(N, TRM_COLS)), dim3(256),
+                       0, s, L, ldl, X, ldx, v, N);
+}
+
 void launch_potrf32(double *A, int64_t lda, int nb, hipStream_t s) {
     if (nb <= 0) return;
     hipLaunchKernelGGL(k_potrf32, dim3(1), dim3(256), 0, s, A, lda, nb);
