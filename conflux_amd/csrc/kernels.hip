@@ -319,49 +319,16 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
             rows[q][tid][cc] = (r < m) ? panel[(int64_t)r * ldp + cc] : 0.0;
     }
 
-    for (int c = 0; c < nb; ++c) {
-        const unsigned int epoch = epoch0 + (unsigned)c;
-        const int par = (int)(epoch & 1u);  // slab parity slot
-        const int cand_off = (int)offsetof(PanelSync2, cand_row) +
-                             (par * CONFLUX_PANEL_MAX_BLOCKS + bid) * PANEL_NB * 8;
-        const int diag_off =
-            (int)offsetof(PanelSync2, diag_row) + par * PANEL_NB * 8;
-        // ---- local candidate: first-max over own rows >= c --------------
-        double amax = -1.0;
-        int arow = m;
-        for (int q = 0; q < 2; ++q) {
-            const int r = q ? r1 : r0;
-            if (r >= c && r < m) {
-                const double a = fabs(rows[q][tid][c]);
-                if (a > amax || (a == amax && r < arow)) { amax = a; arow = r; }
-            }
-        }
-        // barrier-free block argmax: per-wave shuffle tree, 4 partials in
-        // LDS, every thread combines them (redundantly, no 2nd barrier)
-        for (int w = 32; w > 0; w >>= 1) {
-            const double oa = __shfl_down(amax, w);
-            const int orr = __shfl_down(arow, w);
-            if (oa > amax || (oa == amax && orr < arow)) { amax = oa; arow = orr; }
-        }
-        if ((tid & 63) == 0) {
-            red_abs[tid >> 6] = amax;
-            red_row[tid >> 6] = arow;
-        }
-        __syncthreads();
-        double wa = red_abs[0];
-        int wrow = red_row[0];
-        for (int wv = 1; wv < PANEL_TPB / 64; ++wv) {
-            const double oa = red_abs[wv];
-            const int orr = red_row[wv];
-            if (oa > wa || (oa == wa && orr < wrow)) { wa = oa; wrow = orr; }
-        }
-        // ---- cooperative publication: lanes 0-15 stream the winner row,
-        // lanes 16-31 the diag row, as 16-byte sc1 buffer stores.  The
-        // owning (thread, q) of any row r is computable from r directly
-        // (r = bid*RPB + tid + q*TPB), so no extra LDS handoff or barrier:
-        // the publishing lanes read the owner's LDS slot (fresh — last
-        // written before the column loop's closing barrier).  All
-        // publishing lanes are wave 0; one wave drain covers them.
+    // ---- cooperative publication of column pc's candidate: lanes 0-15
+    // stream the winner row, lanes 16-31 the diag row, as 16-byte sc1
+    // buffer stores.  The owning (thread, q) of any row r is computable
+    // from r directly (r = bid*RPB + tid + q*TPB), so the publishing lanes
+    // read the owner's LDS slot; all publishing lanes are wave 0, one wave
+    // drain covers them, then lane 0 posts the key granule (R2: the flag
+    // carries the row) and — if this block owns row pc — the diag flag.
+    auto publish_col = [&](int pc, double wa, int wrow) {
+        const unsigned int pep = epoch0 + (unsigned)pc;
+        const int ppar = (int)(pep & 1u);
         if (tid < 16) {
             const int lrow = (wrow < m) ? wrow - bid * PANEL_RPB : 0;
             const int wq = lrow >= PANEL_TPB;
@@ -371,11 +338,16 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
             x.d[1] = rows[wq][wtid][2 * tid + 1];
             if (2 * tid < nb)
                 __builtin_amdgcn_raw_buffer_store_b128(
-                    x.v, srsrc, cand_off + 16 * tid, 0, /*sc1*/ 16);
-        } else if (tid < 32 && c >= bid * PANEL_RPB &&
-                   c < (bid + 1) * PANEL_RPB) {
+                    x.v, srsrc,
+                    (int)offsetof(PanelSync2, cand_row) +
+                        (ppar * CONFLUX_PANEL_MAX_BLOCKS + bid) * PANEL_NB *
+                            8 +
+                        16 * tid,
+                    0, /*sc1*/ 16);
+        } else if (tid < 32 && pc >= bid * PANEL_RPB &&
+                   pc < (bid + 1) * PANEL_RPB) {
             const int l = tid - 16;
-            const int lrow = c - bid * PANEL_RPB;
+            const int lrow = pc - bid * PANEL_RPB;
             const int dq = lrow >= PANEL_TPB;
             const int dtid = lrow - dq * PANEL_TPB;
             F64x2Bits x;
@@ -383,22 +355,77 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
             x.d[1] = rows[dq][dtid][2 * l + 1];
             if (2 * l < nb)
                 __builtin_amdgcn_raw_buffer_store_b128(
-                    x.v, srsrc, diag_off + 16 * l, 0, /*sc1*/ 16);
+                    x.v, srsrc,
+                    (int)offsetof(PanelSync2, diag_row) +
+                        ppar * PANEL_NB * 8 + 16 * l,
+                    0, /*sc1*/ 16);
         }
         if (tid < 64) drain_stores();
         if (tid == 0) {
             union { double d; unsigned long long u; } a;
             a.d = wa;
-            st_rlx_u64(&sync->key_abs[par][bid], a.u);
+            st_rlx_u64(&sync->key_abs[ppar][bid], a.u);
             drain_stores();
-            // flag granule carries the row (R2: data IS the tag)
-            st_rlx_u64(&sync->key_flag[par][bid],
-                       ((unsigned long long)epoch << 32) | (unsigned)wrow);
-            // the block owning row c raises diag_flag (its storing wave
-            // drained before the barrier above)
-            if (c >= bid * PANEL_RPB && c < (bid + 1) * PANEL_RPB)
-                st_rlx_u32(&sync->diag_flag[par], epoch);
+            st_rlx_u64(&sync->key_flag[ppar][bid],
+                       ((unsigned long long)pep << 32) | (unsigned)wrow);
+            if (pc >= bid * PANEL_RPB && pc < (bid + 1) * PANEL_RPB)
+                st_rlx_u32(&sync->diag_flag[ppar], pep);
         }
+    };
+    // combine the four per-wave partials (redundantly on wave 0's lanes)
+    auto combine_partials = [&](double &wa, int &wrow) {
+        wa = red_abs[0];
+        wrow = red_row[0];
+        for (int wv = 1; wv < PANEL_TPB / 64; ++wv) {
+            const double oa = red_abs[wv];
+            const int orr = red_row[wv];
+            if (oa > wa || (oa == wa && orr < wrow)) { wa = oa; wrow = orr; }
+        }
+    };
+    // per-wave shuffle-tree + LDS partial write of a (|value|, row) pair
+    auto reduce_to_partials = [&](double amax, int arow) {
+        for (int w = 32; w > 0; w >>= 1) {
+            const double oa = __shfl_down(amax, w);
+            const int orr = __shfl_down(arow, w);
+            if (oa > amax || (oa == amax && orr < arow)) {
+                amax = oa;
+                arow = orr;
+            }
+        }
+        if ((tid & 63) == 0) {
+            red_abs[tid >> 6] = amax;
+            red_row[tid >> 6] = arow;
+        }
+    };
+
+    // column 0's candidate: first-max over own rows, then publish.  Every
+    // later column's candidate is reduced and published at the TAIL of the
+    // previous column's update (the values are final there), so a column
+    // step starts directly at the key poll — one barrier and one LDS read
+    // pass fewer per column.
+    {
+        double amax = -1.0;
+        int arow = m;
+        for (int q = 0; q < 2; ++q) {
+            const int r = q ? r1 : r0;
+            if (r >= 0 && r < m) {
+                const double a = fabs(rows[q][tid][0]);
+                if (a > amax || (a == amax && r < arow)) { amax = a; arow = r; }
+            }
+        }
+        reduce_to_partials(amax, arow);
+        __syncthreads();
+        double wa;
+        int wrow;
+        combine_partials(wa, wrow);
+        publish_col(0, wa, wrow);
+    }
+
+    for (int c = 0; c < nb; ++c) {
+        const unsigned int epoch = epoch0 + (unsigned)c;
+        const int par = (int)(epoch & 1u);  // slab parity slot
+        const int diag_off =
+            (int)offsetof(PanelSync2, diag_row) + par * PANEL_NB * 8;
 
         // ---- EVERY block reduces the global winner itself (redundant,
         // deterministic): lane b polls block b's flag granule, reads its
@@ -494,7 +521,9 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         const bool do_scale = pivval != 0.0;
         const double recip = do_scale ? 1.0 / pivval : 0.0;
 
-        // ---- update own rows ---------------------------------------------
+        // ---- update own rows; track the next column's candidate inline --
+        double namax = -1.0;
+        int narow = m;
         for (int q = 0; q < 2; ++q) {
             const int r = q ? r1 : r0;
             if (r >= m) continue;
@@ -514,8 +543,26 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                 for (int cc = c + 1; cc < nb; ++cc)
                     my[cc] -= l * piv_lds[cc];
             }
+            // candidates for column c+1 are exactly the rows updated above
+            if (r > c && c + 1 < nb) {
+                const double a = fabs(my[c + 1]);
+                if (a > namax || (a == namax && r < narow)) {
+                    namax = a;
+                    narow = r;
+                }
+            }
         }
+        if (c + 1 < nb) reduce_to_partials(namax, narow);
         __syncthreads();
+        if (c + 1 < nb) {
+            // tail-publish column c+1's candidate (values final after the
+            // barrier above; the publish overlaps other blocks' update
+            // tails instead of gating the next column's start)
+            double wa;
+            int wrow;
+            combine_partials(wa, wrow);
+            publish_col(c + 1, wa, wrow);
+        }
     }
 
     // write back (plain stores; next kernels see them at the launch boundary)
