@@ -58,7 +58,10 @@ namespace {
 struct RankState {
     int pi = 0, pj = 0, pk = 0, grank = 0;
     // device buffers (fp64 unless noted)
-    double *A11 = nullptr;      // Ml x Nl
+    double *A11 = nullptr;      // Ml x Nl (working copy)
+    double *A11in = nullptr;    // input snapshot (lazily allocated; the
+                                // reference's LU_rep factors a COPY,
+                                // conflux_opt.hpp:398)
     double *A10 = nullptr;      // Ml x v
     double *A01 = nullptr;      // v x Nl
     double *A10Rcv = nullptr;   // Ml x nlayr
@@ -103,6 +106,7 @@ struct Ctx {
     int M = 0, Ml = 0, Nl = 0, Nt = 0, Mt = 0, nlayr = 0, tA11x = 0, tA11y = 0;
     int world = 1, rank = 0;
     bool sim = false;            // all ranks in this process, 1 GPU
+    bool input_dirty = true;     // A11 holds fresh input not yet snapshotted
     bool store_factors = true;
     bool have_comm = false;
     ncclComm_t comm{};
@@ -169,9 +173,9 @@ int ensure_factor_bufs(Ctx &c, RankState &r) {
 }
 
 void free_rank(RankState &r) {
-    for (double *p : {r.A11, r.A10, r.A01, r.A10Rcv, r.A01Rcv, r.A00, r.cand,
-                      r.panel, r.cm, r.A01pack, r.rowtmp, r.redtmp, r.slabs,
-                      r.Fres, r.A10hist})
+    for (double *p : {r.A11, r.A11in, r.A10, r.A01, r.A10Rcv, r.A01Rcv,
+                      r.A00, r.cand, r.panel, r.cm, r.A01pack, r.rowtmp,
+                      r.redtmp, r.slabs, r.Fres, r.A10hist})
         if (p) (void)hipFree(p);
     for (int *p : {r.d_ipiv, r.d_swap, r.d_idx, r.d_gri, r.d_gpivots, r.d_perm})
         if (p) (void)hipFree(p);
@@ -1560,6 +1564,7 @@ int conflux_lu_init_matrix(conflux_lu_ctx *c, uint64_t seed) {
         launch_init_matrix(r.A11, c->Ml, c->Nl, c->v, c->Px, c->Py, r.pi, r.pj,
                            r.pk != 0, seed, c->stream);
     HIPCHK(hipStreamSynchronize(c->stream));
+    c->input_dirty = true;
     return CONFLUX_LU_OK;
 }
 
@@ -1569,14 +1574,18 @@ int conflux_lu_init_matrix_spd(conflux_lu_ctx *c, uint64_t seed) {
         launch_init_matrix_spd(r.A11, c->Ml, c->Nl, c->v, c->Px, c->Py, r.pi,
                                r.pj, r.pk != 0, seed, c->N, c->stream);
     HIPCHK(hipStreamSynchronize(c->stream));
+    c->input_dirty = true;
     return CONFLUX_LU_OK;
 }
+
+static int snapshot_or_restore(conflux_lu_ctx *c);
 
 /* Cholesky factorization A = L L^T of the current (SPD) matrix; the
  * CONFCHOX path (reference src/conflux/cholesky/Cholesky.cpp:857
  * parallelCholesky).  Fres then holds L in the tile-cyclic layout (lower
  * triangle valid). */
 int conflux_chol_factor(conflux_lu_ctx *c, double *elapsed_ms) {
+    if (snapshot_or_restore(c)) return CONFLUX_LU_EHIP;
     int rc = chol_loop(*c, elapsed_ms);
     if (rc)
         std::fprintf(stderr, "[conflux_lu] chol failed: %s\n",
@@ -1594,6 +1603,7 @@ int conflux_lu_set_matrix_local(conflux_lu_ctx *c, const double *local) {
                               hipMemcpyHostToDevice, c->stream));
     }
     HIPCHK(hipStreamSynchronize(c->stream));
+    c->input_dirty = true;
     return CONFLUX_LU_OK;
 }
 
@@ -1610,6 +1620,7 @@ int conflux_lu_set_matrix_sim(conflux_lu_ctx *c, int grank,
         HIPCHK(hipMemcpyAsync(r.A11, local, i64(c->Ml) * c->Nl * 8,
                               hipMemcpyHostToDevice, c->stream));
     HIPCHK(hipStreamSynchronize(c->stream));
+    c->input_dirty = true;
     return CONFLUX_LU_OK;
 }
 
@@ -1633,11 +1644,29 @@ int conflux_lu_store_factors(conflux_lu_ctx *c, int enable) {
     return CONFLUX_LU_OK;
 }
 
+// The reference's LU_rep factors a COPY — lu_params::data survives the call
+// (conflux_opt.hpp:398).  Same semantics here: the first factor() after a
+// matrix upload snapshots A11; later factor() calls restore the snapshot.
+// Both copies run outside the timed region.
+static int snapshot_or_restore(conflux_lu_ctx *c) {
+    for (auto &r : c->rs)
+        if (!r.A11in)
+            HIPCHK(hipMalloc(&r.A11in, i64(c->Ml) * c->Nl * 8));
+    for (auto &r : c->rs) {
+        if (c->input_dirty)
+            HIPCHK(hipMemcpyAsync(r.A11in, r.A11, i64(c->Ml) * c->Nl * 8,
+                                  hipMemcpyDeviceToDevice, c->stream));
+        else
+            HIPCHK(hipMemcpyAsync(r.A11, r.A11in, i64(c->Ml) * c->Nl * 8,
+                                  hipMemcpyDeviceToDevice, c->stream));
+    }
+    HIPCHK(hipStreamSynchronize(c->stream));
+    c->input_dirty = false;
+    return CONFLUX_LU_OK;
+}
+
 int conflux_lu_factor(conflux_lu_ctx *c, double *elapsed_ms) {
-    // factor a COPY: the reference's LU_rep does not clobber lu_params::data
-    // (conflux_opt.hpp:398) — here A11 is the working copy and the caller's
-    // input was uploaded to it; re-upload is the caller's job between reps
-    // (miniapp convention: InitMatrix before each rep, conflux_miniapp:141).
+    if (snapshot_or_restore(c)) return CONFLUX_LU_EHIP;
     int rc = factor_loop(*c, elapsed_ms);
     if (rc) {
         std::fprintf(stderr, "[conflux_lu] factor failed: %s\n",

@@ -173,6 +173,26 @@ def test_lu_single_rank_medium(eng, N, v):
     assert np.array_equal(perm, ref)
 
 
+
+def test_lu_factor_preserves_input(eng):
+    """factor() factors a COPY of the uploaded matrix (the reference's
+    LU_rep does not clobber lu_params::data, conflux_opt.hpp:398): a second
+    factor() without re-upload must reproduce the first result exactly."""
+    N, v = 256, 64
+    A = gen_matrix(N)
+    with eng.Engine(N, v, 1, 1, 1, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor()
+        perm1 = e.get_perm().copy()
+        F1 = e.get_F_global().copy()
+        e.factor()  # no re-upload
+        perm2 = e.get_perm()
+        F2 = e.get_F_global()
+    assert np.array_equal(perm1, perm2)
+    assert np.array_equal(F1, F2)
+
+
 # ---------------- Cholesky (CONFCHOX path, SURVEY §8f1) ---------------------
 
 def _spd(N):
