@@ -154,6 +154,13 @@ class Engine:
         _chk(lib().conflux_chol_factor(self._h, ctypes.byref(ms)), "chol")
         return ms.value
 
+    def validate(self):
+        """Device-side ||PA - LU||_F / ||A||_F of the last factorization
+        (conflux_lu_validate; sim / single-process only)."""
+        r = ctypes.c_double()
+        _chk(lib().conflux_lu_validate(self._h, ctypes.byref(r)), "validate")
+        return r.value
+
     def get_perm(self):
         import numpy as np
         perm = np.zeros(self.M, dtype=np.int32)

@@ -1675,6 +1675,70 @@ int conflux_lu_factor(conflux_lu_ctx *c, double *elapsed_ms) {
     return rc;
 }
 
+/* ||PA - LU||_F / ||A||_F computed ON DEVICE from the stored factors and
+ * the input snapshot — the reference's CONFLUX_WITH_VALIDATION check
+ * (conflux_miniapp.cpp:169-507) without ScaLAPACK.  Available
+ * in simulation mode (any grid, all ranks on this GPU) and 1-process runs;
+ * distributed timing runs validate via --sim at the same (N, v, grid). */
+int conflux_lu_validate(conflux_lu_ctx *c, double *resid) {
+    if (!resid) return CONFLUX_LU_EARG;
+    if (!c->sim && c->world > 1) return CONFLUX_LU_EARG;
+    if (!c->store_factors) return CONFLUX_LU_EARG;
+    for (auto &r : c->rs)
+        if ((r.pk == 0 && !r.Fres) || !r.A11in) return CONFLUX_LU_EARG;
+    const int64_t N = c->N;
+    const int v = c->v;
+    double *Ag = nullptr, *Fg = nullptr, *L = nullptr, *U = nullptr,
+           *PA = nullptr, *d_acc = nullptr;
+    int *d_pm = nullptr;
+    auto cleanup = [&]() {
+        for (double *p : {Ag, Fg, L, U, PA, d_acc})
+            if (p) (void)hipFree(p);
+        if (d_pm) (void)hipFree(d_pm);
+    };
+#define VCHK(x)                                                               \
+    if ((x) != hipSuccess) {                                                  \
+        cleanup();                                                            \
+        return CONFLUX_LU_EHIP;                                               \
+    }
+    VCHK(hipMalloc(&Ag, N * N * 8));
+    VCHK(hipMalloc(&Fg, N * N * 8));
+    VCHK(hipMalloc(&L, N * N * 8));
+    VCHK(hipMalloc(&U, N * N * 8));
+    VCHK(hipMalloc(&PA, N * N * 8));
+    VCHK(hipMalloc(&d_acc, 2 * 8));
+    VCHK(hipMalloc(&d_pm, N * 4));
+    VCHK(hipMemsetAsync(d_acc, 0, 16, c->stream));
+    VCHK(hipMemcpyAsync(d_pm, c->pivotInds.data(), N * 4,
+                        hipMemcpyHostToDevice, c->stream));
+    // assemble global A (input snapshot) and F from the tile-cyclic locals
+    const int Nt = (int)(N / v);
+    for (int ti = 0; ti < Nt; ++ti)
+        for (int tj = 0; tj < Nt; ++tj) {
+            RankState *r = get_rs(*c, ti % c->Px, tj % c->Py, 0);
+            if (!r) continue;  // world==1 covers everything (Px=Py=1)
+            const int64_t lo = i64(ti / c->Px) * v * c->Nl + i64(tj / c->Py) * v;
+            launch_copy2d(r->A11in + lo, c->Nl, Ag + i64(ti) * v * N + tj * v,
+                          N, v, v, c->stream);
+            launch_copy2d(r->Fres + lo, c->Nl, Fg + i64(ti) * v * N + tj * v,
+                          N, v, v, c->stream);
+        }
+    launch_row_gather(Ag, N, PA, N, d_pm, (int)N, N, c->stream);
+    launch_tril_unit(Fg, L, N, c->stream);
+    launch_triu(Fg, U, N, c->stream);
+    launch_dgemm_f64(L, N, U, N, PA, N, (int)N, N, (int)N, c->stream);
+    launch_frob2(PA, N * N, d_acc, c->stream);
+    launch_frob2(Ag, N * N, d_acc + 1, c->stream);
+    double acc[2] = {0, 0};
+    VCHK(hipMemcpyAsync(acc, d_acc, 16, hipMemcpyDeviceToHost, c->stream));
+    VCHK(hipStreamSynchronize(c->stream));
+#undef VCHK
+    cleanup();
+    if (!(acc[1] > 0)) return CONFLUX_LU_EINTERNAL;
+    *resid = std::sqrt(acc[0] / acc[1]);
+    return CONFLUX_LU_OK;
+}
+
 int conflux_lu_get_factors(conflux_lu_ctx *c, double *F_local, int *perm) {
     if (c->sim) return conflux_lu_get_factors_sim(c, 0, F_local, perm);
     RankState &r = c->rs[0];

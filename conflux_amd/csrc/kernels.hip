@@ -1628,6 +1628,56 @@ void launch_trsm_left_mfma(const double *L, int64_t ldl, double *X,
                        0, s, L, ldl, X, ldx, v, N);
 }
 
+// validation helpers (the reference's CONFLUX_WITH_VALIDATION
+// ||PA-LU||_F check, conflux_miniapp.cpp:169-507, without ScaLAPACK) --------
+__global__ void k_tril_unit(const double *__restrict__ F, double *__restrict__ L,
+                            int64_t n) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n * n) return;
+    const int64_t r = i / n, c = i % n;
+    L[i] = (r > c) ? F[i] : (r == c ? 1.0 : 0.0);
+}
+
+__global__ void k_triu(const double *__restrict__ F, double *__restrict__ U,
+                       int64_t n) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n * n) return;
+    const int64_t r = i / n, c = i % n;
+    U[i] = (r <= c) ? F[i] : 0.0;
+}
+
+__global__ void k_frob2(const double *__restrict__ A, int64_t n,
+                        double *__restrict__ out) {
+    __shared__ double part[256];
+    const int tid = threadIdx.x;
+    double s = 0;
+    for (int64_t i = (int64_t)blockIdx.x * 256 + tid; i < n;
+         i += (int64_t)gridDim.x * 256)
+        s += A[i] * A[i];
+    part[tid] = s;
+    __syncthreads();
+    for (int w = 128; w > 0; w >>= 1) {
+        if (tid < w) part[tid] += part[tid + w];
+        __syncthreads();
+    }
+    if (tid == 0) (void)atomicAdd(out, part[0]);
+}
+
+void launch_tril_unit(const double *F, double *L, int64_t n, hipStream_t s) {
+    hipLaunchKernelGGL(k_tril_unit, dim3(cdiv64(n * n, 256)), dim3(256), 0, s,
+                       F, L, n);
+}
+void launch_triu(const double *F, double *U, int64_t n, hipStream_t s) {
+    hipLaunchKernelGGL(k_triu, dim3(cdiv64(n * n, 256)), dim3(256), 0, s, F,
+                       U, n);
+}
+void launch_frob2(const double *A, int64_t nelem, double *out, hipStream_t s) {
+    int blocks = (int)cdiv64(nelem, 256 * 16);
+    if (blocks > 4096) blocks = 4096;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_frob2, dim3(blocks), dim3(256), 0, s, A, nelem, out);
+}
+
 void launch_potrf32(double *A, int64_t lda, int nb, hipStream_t s) {
     if (nb <= 0) return;
     hipLaunchKernelGGL(k_potrf32, dim3(1), dim3(256), 0, s, A, lda, nb);

@@ -174,6 +174,25 @@ def test_lu_single_rank_medium(eng, N, v):
 
 
 
+@pytest.mark.parametrize("N,v,Px,Py,Pz", [(256, 64, 1, 1, 1), (512, 64, 2, 2, 1)])
+def test_lu_validate_device(eng, N, v, Px, Py, Pz):
+    """conflux_lu_validate (device-side ||PA-LU||_F/||A||_F)
+    agrees with the numpy residual computed from the gathered factors."""
+    A = gen_matrix(N)
+    with eng.Engine(N, v, Px, Py, Pz, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor()
+        perm = e.get_perm()
+        F = e.get_F_global()
+        r_dev = e.validate()
+    L = np.tril(F, -1) + np.eye(N)
+    U = np.triu(F)
+    r_np = np.linalg.norm(A[perm] - L @ U) / np.linalg.norm(A)
+    assert r_dev < 1e-13
+    assert abs(r_dev - r_np) < 1e-15 + 0.05 * r_np
+
+
 def test_lu_factor_preserves_input(eng):
     """factor() factors a COPY of the uploaded matrix (the reference's
     LU_rep does not clobber lu_params::data, conflux_opt.hpp:398): a second
