@@ -161,6 +161,14 @@ class Engine:
         _chk(lib().conflux_lu_validate(self._h, ctypes.byref(r)), "validate")
         return r.value
 
+    def validate_cholesky(self):
+        """Device-side ||A - L L^T||_F / ||A||_F of the last Cholesky
+        factorization (sim / single-process only)."""
+        r = ctypes.c_double()
+        _chk(lib().conflux_chol_validate(self._h, ctypes.byref(r)),
+             "chol_validate")
+        return r.value
+
     def get_perm(self):
         import numpy as np
         perm = np.zeros(self.M, dtype=np.int32)

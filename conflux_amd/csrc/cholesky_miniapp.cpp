@@ -135,6 +135,14 @@ int main(int argc, char **argv) {
                         N, P, Px, Py, Pz, ms, v, gflops);
         }
     }
+    if (!timing) {
+        // reference CholeskyValidation prints the residual in DEBUG builds
+        // (Cholesky.cpp:738-772); here computed on device
+        double resid = 0;
+        if (conflux_chol_validate(ctx, &resid) == 0)
+            std::printf("relative residual ||A-LL^T||_F/||A||_F = %.3e\n",
+                        resid);
+    }
     conflux_lu_destroy(ctx);
     return 0;
 }

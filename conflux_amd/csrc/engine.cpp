@@ -1739,6 +1739,62 @@ int conflux_lu_validate(conflux_lu_ctx *c, double *resid) {
     return CONFLUX_LU_OK;
 }
 
+/* Cholesky counterpart: ||A - L L^T||_F / ||A||_F on device (the
+ * reference's CholeskyValidation, Cholesky.cpp:738-772 analogue). */
+int conflux_chol_validate(conflux_lu_ctx *c, double *resid) {
+    if (!resid) return CONFLUX_LU_EARG;
+    if (!c->sim && c->world > 1) return CONFLUX_LU_EARG;
+    for (auto &r : c->rs)
+        if ((r.pk == 0 && !r.Fres) || !r.A11in) return CONFLUX_LU_EARG;
+    const int64_t N = c->N;
+    const int v = c->v;
+    double *Ag = nullptr, *L = nullptr, *R = nullptr, *d_acc = nullptr;
+    auto cleanup = [&]() {
+        for (double *p : {Ag, L, R, d_acc})
+            if (p) (void)hipFree(p);
+    };
+#define VCHK(x)                                                               \
+    if ((x) != hipSuccess) {                                                  \
+        cleanup();                                                            \
+        return CONFLUX_LU_EHIP;                                               \
+    }
+    VCHK(hipMalloc(&Ag, N * N * 8));
+    VCHK(hipMalloc(&L, N * N * 8));
+    VCHK(hipMalloc(&R, N * N * 8));
+    VCHK(hipMalloc(&d_acc, 2 * 8));
+    VCHK(hipMemsetAsync(d_acc, 0, 16, c->stream));
+    const int Nt = (int)(N / v);
+    for (int ti = 0; ti < Nt; ++ti)
+        for (int tj = 0; tj < Nt; ++tj) {
+            RankState *r = get_rs(*c, ti % c->Px, tj % c->Py, 0);
+            if (!r) continue;
+            const int64_t lo = i64(ti / c->Px) * v * c->Nl + i64(tj / c->Py) * v;
+            launch_copy2d(r->A11in + lo, c->Nl, Ag + i64(ti) * v * N + tj * v,
+                          N, v, v, c->stream);
+            launch_copy2d(r->Fres + lo, c->Nl, R + i64(ti) * v * N + tj * v,
+                          N, v, v, c->stream);
+        }
+    launch_tril(R, L, N, c->stream);
+    // the generator fills only the lower triangle per rank (reference
+    // CholeskyIO dsyrk 'L'); mirror it so the Frobenius norms are of the
+    // full symmetric matrix: Ag <- tril(Ag) + tril(Ag,-1)^T via R scratch
+    launch_tril(Ag, R, N, c->stream);
+    launch_copy2d(R, N, Ag, N, (int)N, N, c->stream);
+    launch_transpose_add_lower(Ag, N, c->stream);
+    launch_frob2(Ag, N * N, d_acc + 1, c->stream);
+    launch_copy2d(Ag, N, R, N, (int)N, N, c->stream);
+    launch_dgemm_f64_nt(L, N, L, N, R, N, (int)N, N, (int)N, c->stream);
+    launch_frob2(R, N * N, d_acc, c->stream);
+    double acc[2] = {0, 0};
+    VCHK(hipMemcpyAsync(acc, d_acc, 16, hipMemcpyDeviceToHost, c->stream));
+    VCHK(hipStreamSynchronize(c->stream));
+#undef VCHK
+    cleanup();
+    if (!(acc[1] > 0)) return CONFLUX_LU_EINTERNAL;
+    *resid = std::sqrt(acc[0] / acc[1]);
+    return CONFLUX_LU_OK;
+}
+
 int conflux_lu_get_factors(conflux_lu_ctx *c, double *F_local, int *perm) {
     if (c->sim) return conflux_lu_get_factors_sim(c, 0, F_local, perm);
     RankState &r = c->rs[0];

@@ -1638,6 +1638,14 @@ __global__ void k_tril_unit(const double *__restrict__ F, double *__restrict__ L
     L[i] = (r > c) ? F[i] : (r == c ? 1.0 : 0.0);
 }
 
+__global__ void k_tril(const double *__restrict__ F, double *__restrict__ L,
+                       int64_t n) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n * n) return;
+    const int64_t r = i / n, c = i % n;
+    L[i] = (r >= c) ? F[i] : 0.0;
+}
+
 __global__ void k_triu(const double *__restrict__ F, double *__restrict__ U,
                        int64_t n) {
     const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1666,6 +1674,24 @@ __global__ void k_frob2(const double *__restrict__ A, int64_t n,
 void launch_tril_unit(const double *F, double *L, int64_t n, hipStream_t s) {
     hipLaunchKernelGGL(k_tril_unit, dim3(cdiv64(n * n, 256)), dim3(256), 0, s,
                        F, L, n);
+}
+// A <- A + tril(A,-1)^T, i.e. mirror the strict lower triangle up (makes a
+// lower-stored symmetric matrix explicit).  One thread per upper element.
+__global__ void k_sym_mirror_up(double *__restrict__ A, int64_t n) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n * n) return;
+    const int64_t r = i / n, c = i % n;
+    if (r < c) A[i] = A[c * n + r];
+}
+
+void launch_transpose_add_lower(double *A, int64_t n, hipStream_t s) {
+    hipLaunchKernelGGL(k_sym_mirror_up, dim3(cdiv64(n * n, 256)), dim3(256),
+                       0, s, A, n);
+}
+
+void launch_tril(const double *F, double *L, int64_t n, hipStream_t s) {
+    hipLaunchKernelGGL(k_tril, dim3(cdiv64(n * n, 256)), dim3(256), 0, s, F,
+                       L, n);
 }
 void launch_triu(const double *F, double *U, int64_t n, hipStream_t s) {
     hipLaunchKernelGGL(k_triu, dim3(cdiv64(n * n, 256)), dim3(256), 0, s, F,

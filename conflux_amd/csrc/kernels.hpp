@@ -43,6 +43,8 @@ void launch_trsm_right_mfma(const double *U, int64_t ldu, double *X,
 void launch_trsm_left_mfma(const double *L, int64_t ldl, double *X,
                            int64_t ldx, int v, int64_t N, hipStream_t s);
 void launch_tril_unit(const double *F, double *L, int64_t n, hipStream_t s);
+void launch_transpose_add_lower(double *A, int64_t n, hipStream_t s);
+void launch_tril(const double *F, double *L, int64_t n, hipStream_t s);
 void launch_triu(const double *F, double *U, int64_t n, hipStream_t s);
 void launch_frob2(const double *A, int64_t nelem, double *out, hipStream_t s);
 void launch_potrf32(double *A, int64_t lda, int nb, hipStream_t s);

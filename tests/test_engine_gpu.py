@@ -214,6 +214,23 @@ def test_lu_factor_preserves_input(eng):
 
 # ---------------- Cholesky (CONFCHOX path) ---------------------
 
+
+@pytest.mark.parametrize("N,v,Px,Py,Pz", [(256, 64, 1, 1, 1), (512, 64, 2, 2, 1)])
+def test_chol_validate_device(eng, N, v, Px, Py, Pz):
+    """conflux_chol_validate agrees with the numpy residual."""
+    A = _spd(N)
+    with eng.Engine(N, v, Px, Py, Pz, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor_cholesky()
+        F = e.get_F_global()
+        r_dev = e.validate_cholesky()
+    L = np.tril(F)
+    r_np = np.linalg.norm(A - L @ L.T) / np.linalg.norm(A)
+    assert r_dev < 1e-13
+    assert abs(r_dev - r_np) < 1e-15 + 0.05 * r_np
+
+
 def _spd(N):
     G = gen_matrix(N)
     return 0.5 * (G + G.T) + 2.0 * N * np.eye(N)
