@@ -699,13 +699,15 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
                                   c.stream));
         }
         auto push = [&](double *mat, int64_t ld, int64_t cols) -> int {
+            size_t slot;
+            if (ev_begin(c, 3, 0, &slot)) return CONFLUX_LU_EHIP;
             launch_row_gather(mat, ld, r.rowtmp, cols, r.d_idx, cnt, cols,
                               c.stream);
             launch_row_move(mat, ld, mat, ld, r.d_idx + v, r.d_idx + 2 * v,
                             (int)early.size(), cols, c.stream);
             launch_copy2d(r.rowtmp, cols, mat + i64(f) * ld, ld, cnt, cols,
                           c.stream);
-            return 0;
+            return ev_end(c, slot);
         };
         if (push(r.A11, Nl, Nl)) return CONFLUX_LU_EHIP;
         if (push(r.A10, v, v)) return CONFLUX_LU_EHIP;
@@ -1076,20 +1078,26 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
             const char *e = getenv("CONFLUX_GEMM_CAP");
             env_cap = e ? atoi(e) : 432;
         }
-        gcap = env_cap;
-        // Residency guard: the persistent capped GEMM holds its CUs for the
-        // whole update, so if step k+1's panel needs more co-resident
-        // blocks than the cap leaves free (e.g. tall panels at N=32768),
-        // the overlap would serialize anyway — keep the sequential order
-        // and the full-width GEMM instead.
-        int max_rows = 0;
-        for (auto &r : c.rs) max_rows = std::max(max_rows, r.nact);
-        const int nblocks = (max_rows + conflux_panel_rpb() - 1) /
-                            conflux_panel_rpb();
-        const int free_cus = 256 - (gcap + 1) / 2;
-        if (nblocks > free_cus) {
-            async_look = false;
-            gcap = 0;
+        // Only ranks that RUN step k+1's panel factor need the cap (and
+        // the residency guard); ranks outside column k+1 overlap only the
+        // panel chain's comm and keep the full-width GEMM.
+        int panel_rows = 0;
+        for (auto &r : c.rs)
+            if (r.pj == ncol) panel_rows = std::max(panel_rows, r.nact);
+        if (panel_rows > 0) {
+            gcap = env_cap;
+            // Residency guard: the persistent capped GEMM holds its CUs
+            // for the whole update, so if step k+1's panel needs more
+            // co-resident blocks than the cap leaves free (tall panels at
+            // N >= 32768), the overlap would serialize anyway — keep the
+            // sequential order and the full-width GEMM instead.
+            const int nblocks = (panel_rows + conflux_panel_rpb() - 1) /
+                                conflux_panel_rpb();
+            const int free_cus = 256 - (gcap + 1) / 2;
+            if (nblocks > free_cus) {
+                async_look = false;
+                gcap = 0;
+            }
         }
     }
     // (a) the columns step k+1's panel needs, first
