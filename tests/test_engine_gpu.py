@@ -313,3 +313,47 @@ def test_cholesky_parity_reference_golden(eng):
         assert np.abs(np.tril(F) - Lref).max() < 1e-10, f"{tag}"
         checked += 1
     assert checked >= 3
+
+
+# ---------------- drop-in CLI (the §8b boundary) ----------------------------
+
+def _repo_root():
+    import os
+    return os.path.join(os.path.dirname(__file__), "..")
+
+
+def test_miniapp_cli_lu():
+    """conflux_miniapp drop-in CLI: reference flags, _result_ contract line
+    (conflux_miniapp.cpp:119,156-165) and the validation residual."""
+    import os
+    import re
+    import subprocess
+    out = subprocess.run(
+        [os.path.join(_repo_root(), "conflux_amd", "conflux_miniapp"),
+         "-N", "512", "-b", "64", "--p_grid=2,2,1", "--sim", "-r", "1"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr
+    m = re.search(r"_result_ lu,conflux,512,512,4,2x2x1,time,other,\d+,64",
+                  out.stdout)
+    assert m, out.stdout
+    r = re.search(r"relative residual \|\|PA-LU\|\|_F/\|\|A\|\|_F = ([0-9.e+-]+)",
+                  out.stdout)
+    assert r and float(r.group(1)) < 1e-13, out.stdout
+
+
+def test_miniapp_cli_cholesky():
+    import os
+    import re
+    import subprocess
+    out = subprocess.run(
+        [os.path.join(_repo_root(), "conflux_amd", "cholesky_miniapp"),
+         "--dim", "512", "--tile", "64", "--grid", "2,2,1", "--sim",
+         "--run", "1"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr
+    assert re.search(r"_result_ chol,conflux,512,4,2x2x1,time,\d+,64",
+                     out.stdout), out.stdout
+    r = re.search(
+        r"relative residual \|\|A-LL\^T\|\|_F/\|\|A\|\|_F = ([0-9.e+-]+)",
+        out.stdout)
+    assert r and float(r.group(1)) < 1e-13, out.stdout
