@@ -251,7 +251,9 @@ union F64x2Bits {
 };
 
 // ---------------------------------------------------------------------------
-// TRSM diagonal-block solvers (32-wide blocks; updates via k_dgemm_f64)
+// TRSM diagonal-block solvers (32-wide; the v%32!=0 fallback path and the
+// sub-panel solve inside factor_panel — the whole-panel solves use the
+// fused k_trsm_*_mfma kernels below)
 // ---------------------------------------------------------------------------
 // X (nb x N, row-major ld) <- L^{-1} X with L (nb x nb, ld ldl) unit-lower:
 // thread j owns column j; column values kept in registers.
