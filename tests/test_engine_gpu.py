@@ -193,6 +193,26 @@ def test_lu_validate_device(eng, N, v, Px, Py, Pz):
     assert abs(r_dev - r_np) < 1e-15 + 0.05 * r_np
 
 
+def test_lu_zero_pivot_column(eng):
+    """Singular input (a zero column, placed in the LAST tile column —
+    see tests/test_oracle.py for why): the engine must follow LAPACK's
+    zero-pivot convention (no scaling, factorization continues) and stay
+    bit-exact with the oracle's pivot sequence."""
+    N, v = 128, 64
+    A = gen_matrix(N)
+    A[:, 96] = 0.0
+    p = Params(N, v, 1, 1, 1)
+    r = lu_oracle(A, p)
+    with eng.Engine(N, v, 1, 1, 1, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor()
+        perm = e.get_perm()
+        F = e.get_F_global()
+    assert np.array_equal(perm, r["perm"])
+    assert np.allclose(F, r["F"], atol=1e-11, rtol=0)
+
+
 def test_lu_factor_preserves_input(eng):
     """factor() factors a COPY of the uploaded matrix (the reference's
     LU_rep does not clobber lu_params::data, conflux_opt.hpp:398): a second

@@ -68,3 +68,23 @@ def test_residual_random(grid):
     assert residual_check(A, r["perm"], r["F"]) < TOL_RESID
     # every row used exactly once
     assert np.array_equal(np.sort(r["perm"]), np.arange(N))
+
+
+def test_zero_pivot_column_matches_lapack():
+    """A singular input follows LAPACK's dgetrf convention: the zero pivot
+    skips the scaling (dscal is not applied when the pivot is 0) and the
+    factorization continues.  The zero column sits in the LAST tile column:
+    a zero pivot in an earlier panel makes BLOCKED LU ill-defined
+    downstream (the A10*U^-1 TRSM divides by the zero diagonal — true for
+    the reference's cblas_dtrsm and LAPACK's blocked dgetrf alike)."""
+    import numpy as np
+    import scipy.linalg as la
+    N, v = 64, 32
+    A = gen_matrix(N)
+    A[:, 40] = 0.0
+    r = lu_oracle(A, Params(N, v, 1, 1, 1))
+    lu, piv = la.lu_factor(A)
+    ref = np.arange(N)
+    for i, j in enumerate(piv):
+        ref[i], ref[j] = ref[j], ref[i]
+    assert np.array_equal(r["perm"], ref)
