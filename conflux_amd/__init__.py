@@ -47,6 +47,10 @@ def _load():
     lib.conflux_lu_kernel_stats.argtypes = [
         ctypes.c_void_p, ctypes.c_int, ctypes.POINTER(ctypes.c_double),
         ctypes.POINTER(ctypes.c_long), ctypes.POINTER(ctypes.c_double)]
+    lib.conflux_lu_validate.argtypes = [ctypes.c_void_p,
+                                        ctypes.POINTER(ctypes.c_double)]
+    lib.conflux_chol_validate.argtypes = [ctypes.c_void_p,
+                                          ctypes.POINTER(ctypes.c_double)]
     lib.conflux_lu_destroy.argtypes = [ctypes.c_void_p]
     lib.conflux_lu_build_info.restype = ctypes.c_char_p
     lib.conflux_lu_debug_dgemm.argtypes = [
