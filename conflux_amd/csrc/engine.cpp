@@ -28,6 +28,7 @@
 #include <numeric>
 #include <string>
 #include <unordered_map>
+#include <set>
 #include <vector>
 
 #include "../../include/conflux_lu.h"
@@ -202,6 +203,40 @@ int ev_begin(Ctx &c, int cat, double flops, size_t *slot) {
 int ev_end(Ctx &c, size_t slot) {
     HIPCHK(hipEventRecord(c.evs[slot].b, c.stream));
     return 0;
+}
+
+// CONFLUX_NANCHECK=1 debug tracer: sync + sample a device buffer for
+// non-finite values, print tag.  Zero overhead when the env is unset.
+void nanscan(Ctx &c, const char *tag, int k, const double *p, int64_t nelem) {
+    static int on = -1;
+    if (on < 0) {
+        const char *e = getenv("CONFLUX_NANCHECK");
+        on = e ? atoi(e) : 0;
+    }
+    if (!on || !p || nelem <= 0) return;
+    (void)hipStreamSynchronize(c.stream);
+    const int64_t ns = std::min<int64_t>(nelem, 1 << 20);
+    std::vector<double> s(ns);
+    // sample the head and the tail halves
+    (void)hipMemcpy(s.data(), p, (ns / 2) * 8, hipMemcpyDeviceToHost);
+    (void)hipMemcpy(s.data() + ns / 2, p + nelem - (ns - ns / 2),
+                    (ns - ns / 2) * 8, hipMemcpyDeviceToHost);
+    long bad = 0, zeros = 0;
+    double mn = 1e300, mx = 0;
+    for (double x : s) {
+        if (!std::isfinite(x)) {
+            ++bad;
+            continue;
+        }
+        const double a = std::fabs(x);
+        if (a == 0) ++zeros;
+        if (a > mx) mx = a;
+        if (a < mn) mn = a;
+    }
+    std::fprintf(stderr,
+                 "[nanscan] k=%d %-12s bad=%ld/%lld zeros=%ld min=%.3e "
+                 "max=%.3e\n",
+                 k, tag, bad, (long long)ns, zeros, mn, mx);
 }
 
 // ---------------------------------------------------------------------------
@@ -507,6 +542,27 @@ int phase01(Ctx &c, int k, StepPlan &sp) {
         std::vector<int> ipiv;
         if (n > 0) {
             if (factor_panel(c, r, n, ipiv)) return CONFLUX_LU_EINTERNAL;
+            // dgetrf contract: ipiv[i] in [i, n) — catches any kernel-side
+            // pivot corruption before it can index out of bounds below
+            for (int i = 0; i < std::min(v, n); ++i)
+                if (ipiv[i] < i || ipiv[i] >= n) {
+                    // diagnose: scan the panel input for non-finite data
+                    std::vector<double> samp(i64(std::min(n, 4096)) * v);
+                    (void)hipMemcpy(samp.data(), r.panel, samp.size() * 8,
+                                    hipMemcpyDeviceToHost);
+                    long bad = 0;
+                    double a0 = samp[0];
+                    for (double x : samp)
+                        if (!std::isfinite(x)) ++bad;
+                    c.err = "panel ipiv out of range: k=" +
+                            std::to_string(k) + " i=" + std::to_string(i) +
+                            " ipiv=" + std::to_string(ipiv[i]) +
+                            " n=" + std::to_string(n) +
+                            " nonfinite=" + std::to_string(bad) + "/" +
+                            std::to_string(samp.size()) +
+                            " panel[0]=" + std::to_string(a0);
+                    return CONFLUX_LU_EINTERNAL;
+                }
         }
         ipiv_to_perm(ipiv, n, v, perm);
         if (n == 0) std::iota(perm.begin(), perm.end(), 0);
@@ -680,7 +736,13 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         for (int i = f + cnt; i < c.Ml; ++i)
             if (is_piv[i]) late.push_back(i);
         if (early.size() != late.size()) {
-            c.err = "pivot push invariant";
+            std::set<int> uniq(lr.begin(), lr.end());
+            c.err = "pivot push invariant: k=" + std::to_string(k) +
+                    " cnt=" + std::to_string(cnt) +
+                    " uniq=" + std::to_string(uniq.size()) +
+                    " early=" + std::to_string(early.size()) +
+                    " late=" + std::to_string(late.size()) +
+                    " f=" + std::to_string(f);
             return CONFLUX_LU_EINTERNAL;
         }
         // idx layout in d_idx: [0,v) lrows, [v,2v) early, [2v,3v) late,
@@ -879,7 +941,12 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
     // the step-4 TRSM (the spreads below stay ordered on the main stream,
     // C8 overlapping the step-5 solve like the reference's Iscatterv window,
     // conflux_opt.hpp:1424-1615).
-    const bool split_trsm = !c.sim && c.panel_stream;
+    static int split_env = -1;
+    if (split_env < 0) {
+        const char *se = getenv("CONFLUX_SPLIT_TRSM");
+        split_env = se ? atoi(se) : 1;
+    }
+    const bool split_trsm = !c.sim && c.panel_stream && split_env;
     if (split_trsm) {
         HIPCHK(hipEventRecord(c.ev_t3, c.stream));
         HIPCHK(hipStreamWaitEvent(c.panel_stream, c.ev_t3, 0));
@@ -906,11 +973,16 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         c.stream = saved;
     }
 
+    for (auto &r : c.rs)
+        nanscan(c, "A11.postpush", k, r.A11, i64(c.Ml) * Nl);
     // ---- step 4: A10 <- A10 U^-1, slab-split, spread (C8) ------------------
     for (auto &r : c.rs) {
         if (r.pj != kcol || r.pk != 0) continue;
+        nanscan(c, "A10.pre", k, r.A10 + i64(r.fnp) * v, i64(r.nact) * v);
+        nanscan(c, "A00", k, r.A00, i64(v) * v);
         if (trsm_right_upper(c, r, r.A10 + i64(r.fnp) * v, v, r.nact))
             return CONFLUX_LU_EINTERNAL;
+        nanscan(c, "A10.post", k, r.A10 + i64(r.fnp) * v, i64(r.nact) * v);
         if (c.store_factors)
             launch_copy2d(r.A10 + i64(r.fnp) * v, v,
                           r.A10hist + i64(r.fnp) * Nl + loff, Nl, r.nact, v,
@@ -1100,11 +1172,20 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
             }
         }
     }
+    for (auto &r : c.rs) {
+        nanscan(c, "A10Rcv", k, r.A10Rcv, i64(r.nact) * c.nlayr);
+        nanscan(c, "A01Rcv", k, r.A01Rcv, i64(c.nlayr - 1) * Nl + (Nl - loff));
+    }
     // (a) the columns step k+1's panel needs, first
     if (look)
         for (auto &r : c.rs)
             if (r.pj == ncol)
                 if (gemm_piece(r, lnext, v, 0)) return CONFLUX_LU_EHIP;
+    if (look)
+        for (auto &r : c.rs)
+            if (r.pj == ncol)
+                nanscan(c, "A11.a-cols", k,
+                        r.A11 + i64(r.fnp) * Nl + lnext, i64(r.nact) * Nl);
     if (look && !c.sim && c.panel_stream) {
         HIPCHK(hipEventRecord(c.ev_pc, c.stream));
     }
@@ -1571,6 +1652,7 @@ int conflux_lu_init_matrix(conflux_lu_ctx *c, uint64_t seed) {
     for (auto &r : c->rs)
         launch_init_matrix(r.A11, c->Ml, c->Nl, c->v, c->Px, c->Py, r.pi, r.pj,
                            r.pk != 0, seed, c->stream);
+    HIPCHK(hipGetLastError());  // a rejected launch is otherwise silent
     HIPCHK(hipStreamSynchronize(c->stream));
     c->input_dirty = true;
     return CONFLUX_LU_OK;
@@ -1581,6 +1663,7 @@ int conflux_lu_init_matrix_spd(conflux_lu_ctx *c, uint64_t seed) {
     for (auto &r : c->rs)
         launch_init_matrix_spd(r.A11, c->Ml, c->Nl, c->v, c->Px, c->Py, r.pi,
                                r.pj, r.pk != 0, seed, c->N, c->stream);
+    HIPCHK(hipGetLastError());  // a rejected launch is otherwise silent
     HIPCHK(hipStreamSynchronize(c->stream));
     c->input_dirty = true;
     return CONFLUX_LU_OK;

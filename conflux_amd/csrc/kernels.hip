@@ -54,25 +54,34 @@ DEVFN double gen_entry(int64_t gi, int64_t gj, uint64_t seed) {
     return 5.0 + (double)(h >> 11) * (1.0 / 9007199254740992.0);
 }
 
+// Grid-stride: a flat launch over Ml x Nl elements would need 2^32 threads
+// at N=65536, one past HIP's 32-bit total-thread launch limit — the launch
+// is REJECTED SILENTLY and the matrix stays zero (found the hard way; the
+// launchers below cap the grid and every at-risk kernel strides).
 __global__ void k_init_matrix(double *__restrict__ A, int Ml, int Nl, int v,
                               int Px, int Py, int pi, int pj, int zero_layer,
                               uint64_t seed, int spd, int64_t Nglob) {
-    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     const int64_t total = (int64_t)Ml * Nl;
-    if (idx >= total) return;
-    if (zero_layer) { A[idx] = 0.0; return; }
-    const int r = (int)(idx / Nl), c = (int)(idx % Nl);
-    // local (r, c) -> global (i, j): tile-cyclic map (layout.cpp:95-123)
-    const int64_t gi = (int64_t)(r / v * Px + pi) * v + r % v;
-    const int64_t gj = (int64_t)(c / v * Py + pj) * v + c % v;
-    if (spd) {
-        // symmetric positive definite: sym(gen)/1 + 2N on the diagonal
-        double x = 0.5 * (gen_entry(gi, gj, seed) + gen_entry(gj, gi, seed));
-        if (gi == gj) x += 2.0 * (double)Nglob;
-        A[idx] = x;
-        return;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < total; idx += stride) {
+        if (zero_layer) {
+            A[idx] = 0.0;
+            continue;
+        }
+        const int r = (int)(idx / Nl), c = (int)(idx % Nl);
+        // local (r, c) -> global (i, j): tile-cyclic map (layout.cpp:95-123)
+        const int64_t gi = (int64_t)(r / v * Px + pi) * v + r % v;
+        const int64_t gj = (int64_t)(c / v * Py + pj) * v + c % v;
+        if (spd) {
+            // symmetric positive definite: sym(gen)/1 + 2N on the diagonal
+            double x = 0.5 * (gen_entry(gi, gj, seed) + gen_entry(gj, gi, seed));
+            if (gi == gj) x += 2.0 * (double)Nglob;
+            A[idx] = x;
+            continue;
+        }
+        A[idx] = gen_entry(gi, gj, seed);
     }
-    A[idx] = gen_entry(gi, gj, seed);
 }
 
 // ---------------------------------------------------------------------------
@@ -1262,11 +1271,17 @@ using namespace ck;
 
 static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
+static inline unsigned cap_grid(int64_t total) {
+    int64_t g = cdiv64(total, 256);
+    if (g > (1 << 22)) g = 1 << 22;  // grid-stride kernels cover the rest
+    return (unsigned)g;
+}
+
 void launch_init_matrix(double *A, int Ml, int Nl, int v, int Px, int Py,
                         int pi, int pj, int zero_layer, uint64_t seed,
                         hipStream_t s) {
     const int64_t total = (int64_t)Ml * Nl;
-    hipLaunchKernelGGL(k_init_matrix, dim3(cdiv64(total, 256)), dim3(256), 0, s,
+    hipLaunchKernelGGL(k_init_matrix, dim3(cap_grid(total)), dim3(256), 0, s,
                        A, Ml, Nl, v, Px, Py, pi, pj, zero_layer, seed, 0,
                        (int64_t)0);
 }
@@ -1275,7 +1290,7 @@ void launch_init_matrix_spd(double *A, int Ml, int Nl, int v, int Px, int Py,
                             int pi, int pj, int zero_layer, uint64_t seed,
                             int64_t Nglob, hipStream_t s) {
     const int64_t total = (int64_t)Ml * Nl;
-    hipLaunchKernelGGL(k_init_matrix, dim3(cdiv64(total, 256)), dim3(256), 0, s,
+    hipLaunchKernelGGL(k_init_matrix, dim3(cap_grid(total)), dim3(256), 0, s,
                        A, Ml, Nl, v, Px, Py, pi, pj, zero_layer, seed, 1,
                        Nglob);
 }
@@ -1634,26 +1649,32 @@ void launch_trsm_left_mfma(const double *L, int64_t ldl, double *X,
 // ||PA-LU||_F check, conflux_miniapp.cpp:169-507, without ScaLAPACK) --------
 __global__ void k_tril_unit(const double *__restrict__ F, double *__restrict__ L,
                             int64_t n) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n * n) return;
-    const int64_t r = i / n, c = i % n;
-    L[i] = (r > c) ? F[i] : (r == c ? 1.0 : 0.0);
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n * n; i += stride) {
+        const int64_t r = i / n, c = i % n;
+        L[i] = (r > c) ? F[i] : (r == c ? 1.0 : 0.0);
+    }
 }
 
 __global__ void k_tril(const double *__restrict__ F, double *__restrict__ L,
                        int64_t n) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n * n) return;
-    const int64_t r = i / n, c = i % n;
-    L[i] = (r >= c) ? F[i] : 0.0;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n * n; i += stride) {
+        const int64_t r = i / n, c = i % n;
+        L[i] = (r >= c) ? F[i] : 0.0;
+    }
 }
 
 __global__ void k_triu(const double *__restrict__ F, double *__restrict__ U,
                        int64_t n) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n * n) return;
-    const int64_t r = i / n, c = i % n;
-    U[i] = (r <= c) ? F[i] : 0.0;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n * n; i += stride) {
+        const int64_t r = i / n, c = i % n;
+        U[i] = (r <= c) ? F[i] : 0.0;
+    }
 }
 
 __global__ void k_frob2(const double *__restrict__ A, int64_t n,
@@ -1674,29 +1695,31 @@ __global__ void k_frob2(const double *__restrict__ A, int64_t n,
 }
 
 void launch_tril_unit(const double *F, double *L, int64_t n, hipStream_t s) {
-    hipLaunchKernelGGL(k_tril_unit, dim3(cdiv64(n * n, 256)), dim3(256), 0, s,
+    hipLaunchKernelGGL(k_tril_unit, dim3(cap_grid(n * n)), dim3(256), 0, s,
                        F, L, n);
 }
 // A <- A + tril(A,-1)^T, i.e. mirror the strict lower triangle up (makes a
 // lower-stored symmetric matrix explicit).  One thread per upper element.
 __global__ void k_sym_mirror_up(double *__restrict__ A, int64_t n) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n * n) return;
-    const int64_t r = i / n, c = i % n;
-    if (r < c) A[i] = A[c * n + r];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n * n; i += stride) {
+        const int64_t r = i / n, c = i % n;
+        if (r < c) A[i] = A[c * n + r];
+    }
 }
 
 void launch_transpose_add_lower(double *A, int64_t n, hipStream_t s) {
-    hipLaunchKernelGGL(k_sym_mirror_up, dim3(cdiv64(n * n, 256)), dim3(256),
+    hipLaunchKernelGGL(k_sym_mirror_up, dim3(cap_grid(n * n)), dim3(256),
                        0, s, A, n);
 }
 
 void launch_tril(const double *F, double *L, int64_t n, hipStream_t s) {
-    hipLaunchKernelGGL(k_tril, dim3(cdiv64(n * n, 256)), dim3(256), 0, s, F,
+    hipLaunchKernelGGL(k_tril, dim3(cap_grid(n * n)), dim3(256), 0, s, F,
                        L, n);
 }
 void launch_triu(const double *F, double *U, int64_t n, hipStream_t s) {
-    hipLaunchKernelGGL(k_triu, dim3(cdiv64(n * n, 256)), dim3(256), 0, s, F,
+    hipLaunchKernelGGL(k_triu, dim3(cap_grid(n * n)), dim3(256), 0, s, F,
                        U, n);
 }
 void launch_frob2(const double *A, int64_t nelem, double *out, hipStream_t s) {

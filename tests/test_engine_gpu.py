@@ -377,3 +377,25 @@ def test_miniapp_cli_cholesky():
         r"relative residual \|\|A-LL\^T\|\|_F/\|\|A\|\|_F = ([0-9.e+-]+)",
         out.stdout)
     assert r and float(r.group(1)) < 1e-13, out.stdout
+
+
+@pytest.mark.timeout(600)
+def test_miniapp_cli_full_cfg4_single_gpu():
+    """Regression for the silent 2^32-thread launch rejection: at N=65536
+    a flat per-element launch needs exactly 2^32 threads, one past HIP's
+    limit — the init kernel was rejected silently and LU factored a zero
+    matrix into NaNs.  The grid-stride kernels must handle the full cfg-4
+    size on one GPU (it fits: ~70 GB of 288 GB HBM)."""
+    import os
+    import re
+    import subprocess
+    out = subprocess.run(
+        [os.path.join(_repo_root(), "conflux_amd", "conflux_miniapp"),
+         "-N", "65536", "-b", "512", "--p_grid=1,1,1", "-r", "1",
+         "--timing"],
+        capture_output=True, text=True, timeout=550)
+    assert out.returncode == 0, (out.stdout, out.stderr)
+    m = re.search(r"_result_ lu,conflux,65536,65536,1,1x1x1,time,other,(\d+),512",
+                  out.stdout)
+    assert m, out.stdout
+    assert int(m.group(1)) < 60000  # sanity: minutes would mean spin stalls
