@@ -8,18 +8,7 @@
 // Multi-rank runs self-spawn one process per GPU (like conflux_miniapp);
 // --sim runs all grid ranks on one GPU.  Tile-size default heuristic
 // mirrors the reference (Cholesky.cpp:115-134).
-#include <hip/hip_runtime.h>
-#include <sys/wait.h>
-#include <unistd.h>
-
-#include <cmath>
-#include <cstdio>
-#include <cstdlib>
-#include <cstring>
-#include <string>
-#include <vector>
-
-#include "../../include/conflux_lu.h"
+#include "selfspawn.hpp"
 
 int main(int argc, char **argv) {
     int N = 65536, v = 0, runs = 5, Px = 0, Py = 0, Pz = 0;
@@ -50,7 +39,14 @@ int main(int argc, char **argv) {
             return 1;
         }
     }
-    if (Px <= 0 || Py <= 0 || Pz <= 0) { Px = Py = Pz = 1; }
+    if (Px <= 0 || Py <= 0 || Pz <= 0) {
+        // no grid given: derive from the process count (reference heuristic)
+        int P = 0;
+        if (const char *w = std::getenv("CONFLUX_WORLD")) P = std::atoi(w);
+        if (P <= 0 && !sim) P = conflux_probe_gpu_count();
+        if (P <= 0) P = 1;
+        conflux_grid_from_P(P, &Px, &Py, &Pz);
+    }
     if (v == 0) {
         // reference tile heuristic (Cholesky.cpp:115-134)
         const double ratio = ((double)N * N * Pz / (Px * Py * Pz)) / 1e6;
@@ -65,51 +61,12 @@ int main(int argc, char **argv) {
     if (!sim && P > 1) {
         if (const char *er = std::getenv("CONFLUX_RANK")) {
             rank = std::atoi(er);
-            const char *uf = std::getenv("CONFLUX_UID_FILE");
-            FILE *f = uf ? std::fopen(uf, "rb") : nullptr;
-            if (!f || std::fread(uid, 1, sizeof uid, f) != sizeof uid) return 1;
-            std::fclose(f);
+            if (conflux_resolve_uid(uid)) return 1;
         } else {
-            int ndev = 0;
-            (void)hipGetDeviceCount(&ndev);
-            if (ndev < P) {
-                std::fprintf(stderr, "%d GPUs for grid of %d (use --sim)\n",
-                             ndev, P);
-                return 1;
-            }
-            std::vector<int> pipes(2 * P);
-            for (int r = 0; r < P; ++r)
-                if (pipe(&pipes[2 * r])) return 1;
-            std::vector<pid_t> pids(P);
-            pid_t child = 0;
-            for (int r = 0; r < P; ++r) {
-                pid_t pid = fork();
-                if (pid == 0) {
-                    (void)hipSetDevice(r);
-                    if (r == 0) {
-                        if (conflux_lu_make_uid(uid)) _exit(1);
-                        for (int q = 1; q < P; ++q)
-                            if (write(pipes[2 * q + 1], uid, sizeof uid) !=
-                                (ssize_t)sizeof uid)
-                                _exit(1);
-                    } else if (read(pipes[2 * r], uid, sizeof uid) !=
-                               (ssize_t)sizeof uid) {
-                        _exit(1);
-                    }
-                    rank = r;
-                    child = 1;
-                    break;
-                }
-                pids[r] = pid;
-            }
-            if (!child) {
-                int status = 0, bad = 0;
-                for (int r = 0; r < P; ++r) {
-                    waitpid(pids[r], &status, 0);
-                    if (!WIFEXITED(status) || WEXITSTATUS(status)) bad = 1;
-                }
-                return bad;
-            }
+            conflux_selfspawn(P, argc, argv,
+                              "--grid=" + std::to_string(Px) + "," +
+                                  std::to_string(Py) + "," +
+                                  std::to_string(Pz));
         }
     }
 
@@ -139,7 +96,7 @@ int main(int argc, char **argv) {
         // reference CholeskyValidation prints the residual in DEBUG builds
         // (Cholesky.cpp:738-772); here computed on device
         double resid = 0;
-        if (conflux_chol_validate(ctx, &resid) == 0)
+        if (conflux_chol_validate(ctx, &resid) == 0 && print0)
             std::printf("relative residual ||A-LL^T||_F/||A||_F = %.3e\n",
                         resid);
     }

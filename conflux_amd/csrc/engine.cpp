@@ -165,11 +165,10 @@ int alloc_rank(Ctx &c, RankState &r, int pi, int pj, int pk) {
     return 0;
 }
 
-int ensure_factor_bufs(Ctx &c, RankState &r) {
-    if (!r.Fres) {
-        HIPCHK(hipMalloc(&r.Fres, i64(c.Ml) * c.Nl * 8));
+int ensure_factor_bufs(Ctx &c, RankState &r, bool need_hist = true) {
+    if (!r.Fres) HIPCHK(hipMalloc(&r.Fres, i64(c.Ml) * c.Nl * 8));
+    if (need_hist && !r.A10hist)  // LU-only L-history (chol never reads it)
         HIPCHK(hipMalloc(&r.A10hist, i64(c.Ml) * c.Nl * 8));
-    }
     return 0;
 }
 
@@ -1506,7 +1505,8 @@ int chol_step(Ctx &c, int k) {
 int chol_loop(Ctx &c, double *elapsed_ms) {
     for (auto &r : c.rs) {
         if (c.store_factors) {
-            if (ensure_factor_bufs(c, r)) return CONFLUX_LU_EHIP;
+            if (ensure_factor_bufs(c, r, /*need_hist=*/false))
+                return CONFLUX_LU_EHIP;
             launch_zero2d(r.Fres, c.Nl, c.Ml, c.Nl, c.stream);
         }
     }
@@ -1766,24 +1766,59 @@ int conflux_lu_factor(conflux_lu_ctx *c, double *elapsed_ms) {
     return rc;
 }
 
-/* ||PA - LU||_F / ||A||_F computed ON DEVICE from the stored factors and
- * the input snapshot — the reference's CONFLUX_WITH_VALIDATION check
- * (conflux_miniapp.cpp:169-507) without ScaLAPACK.  Available
- * in simulation mode (any grid, all ranks on this GPU) and 1-process runs;
- * distributed timing runs validate via --sim at the same (N, v, grid). */
-int conflux_lu_validate(conflux_lu_ctx *c, double *resid) {
+/* Validation: ||PA - LU||_F / ||A||_F (or ||A - L L^T||_F /
+ * ||A||_F) computed ON DEVICE from the stored factors and the input
+ * snapshot — the reference's CONFLUX_WITH_VALIDATION check
+ * (conflux_miniapp.cpp:169-507) without ScaLAPACK.
+ *
+ * Stripe-streamed: global F is materialized once (N^2 doubles) and consumed
+ * v rows at a time (the PA / L / residual stripes are v x N), so the
+ * bench-scale single-GPU sizes (N = 65536: ~130 GB of engine state) fit
+ * beside one 32 GB global buffer in 288 GB HBM.  Streaming order makes the
+ * in-place triu legal: stripe i0's GEMM reads U rows k < i0+v only, and all
+ * of those were already stripped of their L half.
+ *
+ * Distributed (world > 1): COLLECTIVE over c->comm — every rank must call.
+ * pk == 0 ranks ship their A11in/Fres locals to grank 0, which assembles
+ * the globals, computes, and broadcasts the residual to all ranks. */
+static int validate_common(conflux_lu_ctx *c, bool chol, double *resid) {
     if (!resid) return CONFLUX_LU_EARG;
-    if (!c->sim && c->world > 1) return CONFLUX_LU_EARG;
     if (!c->store_factors) return CONFLUX_LU_EARG;
     for (auto &r : c->rs)
         if ((r.pk == 0 && !r.Fres) || !r.A11in) return CONFLUX_LU_EARG;
     const int64_t N = c->N;
     const int v = c->v;
-    double *Ag = nullptr, *Fg = nullptr, *L = nullptr, *U = nullptr,
-           *PA = nullptr, *d_acc = nullptr;
+    const bool dist = !c->sim && c->world > 1;
+    RankState &me = c->rs[0];
+
+    if (dist && me.grank != 0) {
+        // non-root: contribute tiles (pk == 0 layers hold the data), then
+        // receive the residual broadcast
+        if (me.pk == 0) {
+            NCCLCHK(ncclSend(me.A11in, i64(c->Ml) * c->Nl, ncclDouble, 0,
+                             c->comm, c->stream));
+            NCCLCHK(ncclSend(me.Fres, i64(c->Ml) * c->Nl, ncclDouble, 0,
+                             c->comm, c->stream));
+        }
+        double *d_r = nullptr;
+        HIPCHK(hipMalloc(&d_r, 8));
+        NCCLCHK(ncclBroadcast(d_r, d_r, 1, ncclDouble, 0, c->comm, c->stream));
+        HIPCHK(hipMemcpyAsync(resid, d_r, 8, hipMemcpyDeviceToHost, c->stream));
+        HIPCHK(hipStreamSynchronize(c->stream));
+        (void)hipFree(d_r);
+        return CONFLUX_LU_OK;
+    }
+
+    // root / single-process path
+    double *Ag = nullptr, *Fg = nullptr, *L = nullptr, *C = nullptr,
+           *d_acc = nullptr, *tmp = nullptr;
     int *d_pm = nullptr;
+    // world==1 LU: A11in IS the global matrix (Px=Py=1) and is only read —
+    // alias it instead of burning another N^2 (the N=65536 case needs this)
+    const bool aliasA = (!chol && !c->sim && c->world == 1);
     auto cleanup = [&]() {
-        for (double *p : {Ag, Fg, L, U, PA, d_acc})
+        if (Ag && !aliasA) (void)hipFree(Ag);
+        for (double *p : {Fg, L, C, d_acc, tmp})
             if (p) (void)hipFree(p);
         if (d_pm) (void)hipFree(d_pm);
     };
@@ -1792,98 +1827,143 @@ int conflux_lu_validate(conflux_lu_ctx *c, double *resid) {
         cleanup();                                                            \
         return CONFLUX_LU_EHIP;                                               \
     }
-    VCHK(hipMalloc(&Ag, N * N * 8));
+#define VNCCL(x)                                                              \
+    if ((x) != ncclSuccess) {                                                 \
+        cleanup();                                                            \
+        return CONFLUX_LU_ECOMM;                                              \
+    }
+    if (aliasA)
+        Ag = me.A11in;
+    else
+        VCHK(hipMalloc(&Ag, N * N * 8));
     VCHK(hipMalloc(&Fg, N * N * 8));
-    VCHK(hipMalloc(&L, N * N * 8));
-    VCHK(hipMalloc(&U, N * N * 8));
-    VCHK(hipMalloc(&PA, N * N * 8));
+    if (!chol) {
+        VCHK(hipMalloc(&L, i64(v) * N * 8));
+        VCHK(hipMalloc(&d_pm, N * 4));
+    }
+    VCHK(hipMalloc(&C, i64(v) * N * 8));
     VCHK(hipMalloc(&d_acc, 2 * 8));
-    VCHK(hipMalloc(&d_pm, N * 4));
     VCHK(hipMemsetAsync(d_acc, 0, 16, c->stream));
-    VCHK(hipMemcpyAsync(d_pm, c->pivotInds.data(), N * 4,
-                        hipMemcpyHostToDevice, c->stream));
+
     // assemble global A (input snapshot) and F from the tile-cyclic locals
+    // (owner map layout.cpp:95-123)
     const int Nt = (int)(N / v);
-    for (int ti = 0; ti < Nt; ++ti)
-        for (int tj = 0; tj < Nt; ++tj) {
-            RankState *r = get_rs(*c, ti % c->Px, tj % c->Py, 0);
-            if (!r) continue;  // world==1 covers everything (Px=Py=1)
-            const int64_t lo = i64(ti / c->Px) * v * c->Nl + i64(tj / c->Py) * v;
-            launch_copy2d(r->A11in + lo, c->Nl, Ag + i64(ti) * v * N + tj * v,
-                          N, v, v, c->stream);
-            launch_copy2d(r->Fres + lo, c->Nl, Fg + i64(ti) * v * N + tj * v,
-                          N, v, v, c->stream);
+    auto scatter_tiles = [&](const double *localA, const double *localF,
+                             int pi, int pj) {
+        if (c->Px == 1 && c->Py == 1) {  // local IS global: one bulk copy
+            if (localA && !aliasA)
+                launch_copy2d(localA, c->Nl, Ag, N, c->Ml, c->Nl, c->stream);
+            if (localF)
+                launch_copy2d(localF, c->Nl, Fg, N, c->Ml, c->Nl, c->stream);
+            return;
         }
-    launch_row_gather(Ag, N, PA, N, d_pm, (int)N, N, c->stream);
-    launch_tril_unit(Fg, L, N, c->stream);
-    launch_triu(Fg, U, N, c->stream);
-    launch_dgemm_f64(L, N, U, N, PA, N, (int)N, N, (int)N, c->stream);
-    launch_frob2(PA, N * N, d_acc, c->stream);
-    launch_frob2(Ag, N * N, d_acc + 1, c->stream);
+        for (int ti = pi; ti < Nt; ti += c->Px)
+            for (int tj = pj; tj < Nt; tj += c->Py) {
+                const int64_t lo =
+                    i64(ti / c->Px) * v * c->Nl + i64(tj / c->Py) * v;
+                if (localA && !aliasA)
+                    launch_copy2d(localA + lo, c->Nl,
+                                  Ag + i64(ti) * v * N + tj * v, N, v, v,
+                                  c->stream);
+                if (localF)
+                    launch_copy2d(localF + lo, c->Nl,
+                                  Fg + i64(ti) * v * N + tj * v, N, v, v,
+                                  c->stream);
+            }
+    };
+    if (c->sim) {
+        for (int pi = 0; pi < c->Px; ++pi)
+            for (int pj = 0; pj < c->Py; ++pj) {
+                RankState *r = get_rs(*c, pi, pj, 0);
+                scatter_tiles(r->A11in, r->Fres, pi, pj);
+            }
+    } else if (!dist) {
+        scatter_tiles(me.A11in, me.Fres, 0, 0);
+    } else {
+        VCHK(hipMalloc(&tmp, i64(c->Ml) * c->Nl * 8));
+        for (int pi = 0; pi < c->Px; ++pi)
+            for (int pj = 0; pj < c->Py; ++pj) {
+                if (pi == me.pi && pj == me.pj) {
+                    scatter_tiles(me.A11in, me.Fres, pi, pj);
+                    continue;
+                }
+                const int src = grank_of(*c, pi, pj, 0);
+                VNCCL(ncclRecv(tmp, i64(c->Ml) * c->Nl, ncclDouble, src,
+                               c->comm, c->stream));
+                scatter_tiles(tmp, nullptr, pi, pj);  // A tiles
+                // (scatter_tiles writes F from its 2nd arg; split the call)
+                VNCCL(ncclRecv(tmp, i64(c->Ml) * c->Nl, ncclDouble, src,
+                               c->comm, c->stream));
+                scatter_tiles(nullptr, tmp, pi, pj);  // F tiles
+            }
+    }
+
+    if (!chol) {
+        VCHK(hipMemcpyAsync(d_pm, c->pivotInds.data(), N * 4,
+                            hipMemcpyHostToDevice, c->stream));
+        launch_frob2(Ag, N * N, d_acc + 1, c->stream);
+        for (int64_t i0 = 0; i0 < N; i0 += v) {
+            const int K = (int)(i0 + v);
+            // PA stripe: rows perm[i0 .. i0+v)
+            launch_row_gather(Ag, N, C, N, d_pm + i0, v, N, c->stream);
+            // L stripe (strict lower + unit diag), then strip those rows
+            // to pure U in place — later stripes only read rows < their K
+            launch_tril_unit_rows(Fg + i0 * N, N, L, N, v, i0, K, c->stream);
+            launch_triu_rows(Fg + i0 * N, N, v, i0, c->stream);
+            launch_dgemm_f64(L, N, Fg, N, C, N, v, N, K, c->stream);
+            launch_frob2(C, i64(v) * N, d_acc, c->stream);
+        }
+    } else {
+        // symmetrize from the lower triangle (the generator / reference
+        // CholeskyIO fill lower-stored input, dsyrk 'L')
+        launch_transpose_add_lower(Ag, N, c->stream);
+        launch_frob2(Ag, N * N, d_acc + 1, c->stream);
+        // Fg <- tril(Fg) in place (upper half of the stored L is junk);
+        // the stripe GEMMs then read L directly out of Fg
+        launch_tril(Fg, Fg, N, c->stream);
+        for (int64_t i0 = 0; i0 < N; i0 += v) {
+            const int K = (int)(i0 + v);  // L rows i0.. have cols <= i0+v-1
+            launch_copy2d(Ag + i0 * N, N, C, N, v, N, c->stream);
+            launch_dgemm_f64_nt(Fg + i0 * N, N, Fg, N, C, N, v, N, K,
+                                c->stream);
+            launch_frob2(C, i64(v) * N, d_acc, c->stream);
+        }
+    }
     double acc[2] = {0, 0};
     VCHK(hipMemcpyAsync(acc, d_acc, 16, hipMemcpyDeviceToHost, c->stream));
     VCHK(hipStreamSynchronize(c->stream));
-#undef VCHK
-    cleanup();
-    if (!(acc[1] > 0)) return CONFLUX_LU_EINTERNAL;
+    if (!(acc[1] > 0)) {
+        cleanup();
+        return CONFLUX_LU_EINTERNAL;
+    }
     *resid = std::sqrt(acc[0] / acc[1]);
+    if (dist) {
+        double *d_r = nullptr;
+        VCHK(hipMalloc(&d_r, 8));
+        VCHK(hipMemcpyAsync(d_r, resid, 8, hipMemcpyHostToDevice, c->stream));
+        const ncclResult_t e =
+            ncclBroadcast(d_r, d_r, 1, ncclDouble, 0, c->comm, c->stream);
+        (void)hipStreamSynchronize(c->stream);
+        (void)hipFree(d_r);
+        if (e != ncclSuccess) {
+            cleanup();
+            return CONFLUX_LU_ECOMM;
+        }
+    }
+#undef VCHK
+#undef VNCCL
+    cleanup();
     return CONFLUX_LU_OK;
 }
 
-/* Cholesky counterpart: ||A - L L^T||_F / ||A||_F on device (the
- * reference's CholeskyValidation, Cholesky.cpp:738-772 analogue). */
+int conflux_lu_validate(conflux_lu_ctx *c, double *resid) {
+    return validate_common(c, false, resid);
+}
+
+/* Cholesky counterpart: ||A - L L^T||_F / ||A||_F (the reference's
+ * CholeskyValidation, Cholesky.cpp:738-772 analogue). */
 int conflux_chol_validate(conflux_lu_ctx *c, double *resid) {
-    if (!resid) return CONFLUX_LU_EARG;
-    if (!c->sim && c->world > 1) return CONFLUX_LU_EARG;
-    for (auto &r : c->rs)
-        if ((r.pk == 0 && !r.Fres) || !r.A11in) return CONFLUX_LU_EARG;
-    const int64_t N = c->N;
-    const int v = c->v;
-    double *Ag = nullptr, *L = nullptr, *R = nullptr, *d_acc = nullptr;
-    auto cleanup = [&]() {
-        for (double *p : {Ag, L, R, d_acc})
-            if (p) (void)hipFree(p);
-    };
-#define VCHK(x)                                                               \
-    if ((x) != hipSuccess) {                                                  \
-        cleanup();                                                            \
-        return CONFLUX_LU_EHIP;                                               \
-    }
-    VCHK(hipMalloc(&Ag, N * N * 8));
-    VCHK(hipMalloc(&L, N * N * 8));
-    VCHK(hipMalloc(&R, N * N * 8));
-    VCHK(hipMalloc(&d_acc, 2 * 8));
-    VCHK(hipMemsetAsync(d_acc, 0, 16, c->stream));
-    const int Nt = (int)(N / v);
-    for (int ti = 0; ti < Nt; ++ti)
-        for (int tj = 0; tj < Nt; ++tj) {
-            RankState *r = get_rs(*c, ti % c->Px, tj % c->Py, 0);
-            if (!r) continue;
-            const int64_t lo = i64(ti / c->Px) * v * c->Nl + i64(tj / c->Py) * v;
-            launch_copy2d(r->A11in + lo, c->Nl, Ag + i64(ti) * v * N + tj * v,
-                          N, v, v, c->stream);
-            launch_copy2d(r->Fres + lo, c->Nl, R + i64(ti) * v * N + tj * v,
-                          N, v, v, c->stream);
-        }
-    launch_tril(R, L, N, c->stream);
-    // the generator fills only the lower triangle per rank (reference
-    // CholeskyIO dsyrk 'L'); mirror it so the Frobenius norms are of the
-    // full symmetric matrix: Ag <- tril(Ag) + tril(Ag,-1)^T via R scratch
-    launch_tril(Ag, R, N, c->stream);
-    launch_copy2d(R, N, Ag, N, (int)N, N, c->stream);
-    launch_transpose_add_lower(Ag, N, c->stream);
-    launch_frob2(Ag, N * N, d_acc + 1, c->stream);
-    launch_copy2d(Ag, N, R, N, (int)N, N, c->stream);
-    launch_dgemm_f64_nt(L, N, L, N, R, N, (int)N, N, (int)N, c->stream);
-    launch_frob2(R, N * N, d_acc, c->stream);
-    double acc[2] = {0, 0};
-    VCHK(hipMemcpyAsync(acc, d_acc, 16, hipMemcpyDeviceToHost, c->stream));
-    VCHK(hipStreamSynchronize(c->stream));
-#undef VCHK
-    cleanup();
-    if (!(acc[1] > 0)) return CONFLUX_LU_EINTERNAL;
-    *resid = std::sqrt(acc[0] / acc[1]);
-    return CONFLUX_LU_OK;
+    return validate_common(c, true, resid);
 }
 
 int conflux_lu_get_factors(conflux_lu_ctx *c, double *F_local, int *perm) {
@@ -1978,7 +2058,6 @@ int conflux_lu_debug_getrf(int n, int v, double *panel, int *ipiv_out) {
     HIPCHK(hipMalloc(&r.d_ipiv, (v + 8) * 4));
     HIPCHK(hipMalloc(&r.d_swap, 128 * 4));
     HIPCHK(hipMalloc(&r.rowtmp, i64(64) * v * 8));
-    HIPCHK(hipMalloc(&r.d_swap, 128 * 4));
     HIPCHK(hipMalloc(&r.sync, conflux_panel_sync_bytes()));
     HIPCHK(hipMemset(r.sync, 0, conflux_panel_sync_bytes()));
     HIPCHK(hipMemcpy(r.panel, panel, i64(n) * v * 8, hipMemcpyHostToDevice));

@@ -90,25 +90,29 @@ __global__ void k_init_matrix(double *__restrict__ A, int Ml, int Nl, int v,
 __global__ void k_copy2d(const double *__restrict__ src, int64_t lds,
                          double *__restrict__ dst, int64_t ldd,
                          int rows, int64_t cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)rows * cols) return;
-    const int64_t r = i / cols, c = i % cols;
-    dst[r * ldd + c] = src[r * lds + c];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)rows * cols; i += stride) {
+        const int64_t r = i / cols, c = i % cols;
+        dst[r * ldd + c] = src[r * lds + c];
+    }
 }
 
 __global__ void k_zero2d(double *__restrict__ dst, int64_t ldd, int rows,
                          int64_t cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)rows * cols) return;
-    dst[(i / cols) * ldd + i % cols] = 0.0;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)rows * cols; i += stride)
+        dst[(i / cols) * ldd + i % cols] = 0.0;
 }
 
 __global__ void k_add2d(const double *__restrict__ src, int64_t lds,
                         double *__restrict__ dst, int64_t ldd, int rows,
                         int64_t cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)rows * cols) return;
-    dst[(i / cols) * ldd + i % cols] += src[(i / cols) * lds + i % cols];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)rows * cols; i += stride)
+        dst[(i / cols) * ldd + i % cols] += src[(i / cols) * lds + i % cols];
 }
 
 // dst row i <- src row idx[i]   (gather; push_pivots_up phase 1/3,
@@ -117,10 +121,12 @@ __global__ void k_row_gather(const double *__restrict__ src, int64_t lds,
                              double *__restrict__ dst, int64_t ldd,
                              const int *__restrict__ idx, int n_rows,
                              int64_t cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)n_rows * cols) return;
-    const int64_t r = i / cols, c = i % cols;
-    dst[r * ldd + c] = src[(int64_t)idx[r] * lds + c];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)n_rows * cols; i += stride) {
+        const int64_t r = i / cols, c = i % cols;
+        dst[r * ldd + c] = src[(int64_t)idx[r] * lds + c];
+    }
 }
 
 // dst row idx[i] <- src row i   (scatter; push_pivots_up phase 2,
@@ -129,10 +135,12 @@ __global__ void k_row_scatter(const double *__restrict__ src, int64_t lds,
                               double *__restrict__ dst, int64_t ldd,
                               const int *__restrict__ idx, int n_rows,
                               int64_t cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)n_rows * cols) return;
-    const int64_t r = i / cols, c = i % cols;
-    dst[(int64_t)idx[r] * ldd + c] = src[r * lds + c];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)n_rows * cols; i += stride) {
+        const int64_t r = i / cols, c = i % cols;
+        dst[(int64_t)idx[r] * ldd + c] = src[r * lds + c];
+    }
 }
 
 // src rows idx[i] copied to dst rows dst_idx[i] within SAME buffer is unsafe;
@@ -147,11 +155,13 @@ __global__ void k_rowperm_gather_skip(const double *__restrict__ src,
                                       const int *__restrict__ src_idx,
                                       int row_base, int n_rows, int64_t skip0,
                                       int64_t skipn, int64_t tot_cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)n_rows * tot_cols) return;
-    const int64_t r = i / tot_cols, cc = i % tot_cols;
-    const int64_t c = (cc < skip0) ? cc : cc + skipn;
-    tmp[r * tot_cols + cc] = src[(int64_t)(row_base + src_idx[r]) * lds + c];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)n_rows * tot_cols; i += stride) {
+        const int64_t r = i / tot_cols, cc = i % tot_cols;
+        const int64_t c = (cc < skip0) ? cc : cc + skipn;
+        tmp[r * tot_cols + cc] = src[(int64_t)(row_base + src_idx[r]) * lds + c];
+    }
 }
 
 __global__ void k_rowperm_scatter_skip(const double *__restrict__ tmp,
@@ -159,11 +169,13 @@ __global__ void k_rowperm_scatter_skip(const double *__restrict__ tmp,
                                        const int *__restrict__ dst_idx,
                                        int row_base, int n_rows, int64_t skip0,
                                        int64_t skipn, int64_t tot_cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)n_rows * tot_cols) return;
-    const int64_t r = i / tot_cols, cc = i % tot_cols;
-    const int64_t c = (cc < skip0) ? cc : cc + skipn;
-    dst[(int64_t)(dst_idx[r] + row_base) * ldd + c] = tmp[r * tot_cols + cc];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)n_rows * tot_cols; i += stride) {
+        const int64_t r = i / tot_cols, cc = i % tot_cols;
+        const int64_t c = (cc < skip0) ? cc : cc + skipn;
+        dst[(int64_t)(dst_idx[r] + row_base) * ldd + c] = tmp[r * tot_cols + cc];
+    }
 }
 
 // dst row dst_idx[i] <- src row src_idx[i]; row sets must be disjoint
@@ -173,10 +185,12 @@ __global__ void k_row_move(const double *__restrict__ src, int64_t lds,
                            const int *__restrict__ src_idx,
                            const int *__restrict__ dst_idx, int n_rows,
                            int64_t cols) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= (int64_t)n_rows * cols) return;
-    const int64_t r = i / cols, c = i % cols;
-    dst[(int64_t)dst_idx[r] * ldd + c] = src[(int64_t)src_idx[r] * lds + c];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)n_rows * cols; i += stride) {
+        const int64_t r = i / cols, c = i % cols;
+        dst[(int64_t)dst_idx[r] * ldd + c] = src[(int64_t)src_idx[r] * lds + c];
+    }
 }
 
 
@@ -1298,21 +1312,21 @@ void launch_init_matrix_spd(double *A, int Ml, int Nl, int v, int Px, int Py,
 void launch_copy2d(const double *src, int64_t lds, double *dst, int64_t ldd,
                    int rows, int64_t cols, hipStream_t s) {
     if (rows <= 0 || cols <= 0) return;
-    hipLaunchKernelGGL(k_copy2d, dim3(cdiv64((int64_t)rows * cols, 256)),
+    hipLaunchKernelGGL(k_copy2d, dim3(cap_grid((int64_t)rows * cols)),
                        dim3(256), 0, s, src, lds, dst, ldd, rows, cols);
 }
 
 void launch_zero2d(double *dst, int64_t ldd, int rows, int64_t cols,
                    hipStream_t s) {
     if (rows <= 0 || cols <= 0) return;
-    hipLaunchKernelGGL(k_zero2d, dim3(cdiv64((int64_t)rows * cols, 256)),
+    hipLaunchKernelGGL(k_zero2d, dim3(cap_grid((int64_t)rows * cols)),
                        dim3(256), 0, s, dst, ldd, rows, cols);
 }
 
 void launch_add2d(const double *src, int64_t lds, double *dst, int64_t ldd,
                   int rows, int64_t cols, hipStream_t s) {
     if (rows <= 0 || cols <= 0) return;
-    hipLaunchKernelGGL(k_add2d, dim3(cdiv64((int64_t)rows * cols, 256)),
+    hipLaunchKernelGGL(k_add2d, dim3(cap_grid((int64_t)rows * cols)),
                        dim3(256), 0, s, src, lds, dst, ldd, rows, cols);
 }
 
@@ -1320,7 +1334,7 @@ void launch_row_gather(const double *src, int64_t lds, double *dst,
                        int64_t ldd, const int *idx, int n_rows, int64_t cols,
                        hipStream_t s) {
     if (n_rows <= 0 || cols <= 0) return;
-    hipLaunchKernelGGL(k_row_gather, dim3(cdiv64((int64_t)n_rows * cols, 256)),
+    hipLaunchKernelGGL(k_row_gather, dim3(cap_grid((int64_t)n_rows * cols)),
                        dim3(256), 0, s, src, lds, dst, ldd, idx, n_rows, cols);
 }
 
@@ -1328,7 +1342,7 @@ void launch_row_scatter(const double *src, int64_t lds, double *dst,
                         int64_t ldd, const int *idx, int n_rows, int64_t cols,
                         hipStream_t s) {
     if (n_rows <= 0 || cols <= 0) return;
-    hipLaunchKernelGGL(k_row_scatter, dim3(cdiv64((int64_t)n_rows * cols, 256)),
+    hipLaunchKernelGGL(k_row_scatter, dim3(cap_grid((int64_t)n_rows * cols)),
                        dim3(256), 0, s, src, lds, dst, ldd, idx, n_rows, cols);
 }
 
@@ -1342,10 +1356,10 @@ void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
                          double *tmp, hipStream_t s) {
     if (n_rows <= 0 || tot_cols <= 0) return;
     const int64_t n = (int64_t)n_rows * tot_cols;
-    hipLaunchKernelGGL(k_rowperm_gather_skip, dim3(cdiv64(n, 256)), dim3(256),
+    hipLaunchKernelGGL(k_rowperm_gather_skip, dim3(cap_grid(n)), dim3(256),
                        0, s, mat, ld, tmp, src_idx, row_base, n_rows, skip0,
                        skipn, tot_cols);
-    hipLaunchKernelGGL(k_rowperm_scatter_skip, dim3(cdiv64(n, 256)), dim3(256),
+    hipLaunchKernelGGL(k_rowperm_scatter_skip, dim3(cap_grid(n)), dim3(256),
                        0, s, tmp, mat, ld, dst_idx, row_base, n_rows, skip0,
                        skipn, tot_cols);
 }
@@ -1698,6 +1712,49 @@ void launch_tril_unit(const double *F, double *L, int64_t n, hipStream_t s) {
     hipLaunchKernelGGL(k_tril_unit, dim3(cap_grid(n * n)), dim3(256), 0, s,
                        F, L, n);
 }
+
+// stripe variants for the streamed ||PA-LU|| validation: operate on a
+// `rows`-row horizontal stripe whose first row is GLOBAL row `row0` of the
+// factored matrix, so validation peaks at one N^2 buffer instead of five
+// (bench-scale single-GPU sizes fit in HBM this way).
+__global__ void k_tril_unit_rows(const double *__restrict__ F, int64_t ldf,
+                                 double *__restrict__ L, int64_t ldl,
+                                 int rows, int64_t row0, int64_t ncols) {
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)rows * ncols; i += stride) {
+        const int64_t r = i / ncols, c = i % ncols, g = row0 + r;
+        L[r * ldl + c] = (g > c) ? F[r * ldf + c] : (g == c ? 1.0 : 0.0);
+    }
+}
+
+// zero the strict lower part of stripe rows in place (rows become pure U)
+__global__ void k_triu_rows(double *__restrict__ F, int64_t ldf, int rows,
+                            int64_t row0) {
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t w = row0 + rows;  // strict-lower cols are < global row < w
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)rows * w; i += stride) {
+        const int64_t r = i / w, c = i % w;
+        if (c < row0 + r) F[r * ldf + c] = 0.0;
+    }
+}
+
+void launch_tril_unit_rows(const double *F, int64_t ldf, double *L,
+                           int64_t ldl, int rows, int64_t row0, int64_t ncols,
+                           hipStream_t s) {
+    if (rows <= 0 || ncols <= 0) return;
+    hipLaunchKernelGGL(k_tril_unit_rows, dim3(cap_grid((int64_t)rows * ncols)),
+                       dim3(256), 0, s, F, ldf, L, ldl, rows, row0, ncols);
+}
+
+void launch_triu_rows(double *F, int64_t ldf, int rows, int64_t row0,
+                      hipStream_t s) {
+    if (rows <= 0) return;
+    hipLaunchKernelGGL(k_triu_rows,
+                       dim3(cap_grid((int64_t)rows * (row0 + rows))),
+                       dim3(256), 0, s, F, ldf, rows, row0);
+}
 // A <- A + tril(A,-1)^T, i.e. mirror the strict lower triangle up (makes a
 // lower-stored symmetric matrix explicit).  One thread per upper element.
 __global__ void k_sym_mirror_up(double *__restrict__ A, int64_t n) {
@@ -1783,7 +1840,7 @@ void launch_row_move(const double *src, int64_t lds, double *dst, int64_t ldd,
                      const int *src_idx, const int *dst_idx, int n_rows,
                      int64_t cols, hipStream_t s) {
     if (n_rows <= 0 || cols <= 0) return;
-    hipLaunchKernelGGL(k_row_move, dim3(cdiv64((int64_t)n_rows * cols, 256)),
+    hipLaunchKernelGGL(k_row_move, dim3(cap_grid((int64_t)n_rows * cols)),
                        dim3(256), 0, s, src, lds, dst, ldd, src_idx, dst_idx,
                        n_rows, cols);
 }

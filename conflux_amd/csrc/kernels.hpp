@@ -43,6 +43,11 @@ void launch_trsm_right_mfma(const double *U, int64_t ldu, double *X,
 void launch_trsm_left_mfma(const double *L, int64_t ldl, double *X,
                            int64_t ldx, int v, int64_t N, hipStream_t s);
 void launch_tril_unit(const double *F, double *L, int64_t n, hipStream_t s);
+void launch_tril_unit_rows(const double *F, int64_t ldf, double *L,
+                           int64_t ldl, int rows, int64_t row0, int64_t ncols,
+                           hipStream_t s);
+void launch_triu_rows(double *F, int64_t ldf, int rows, int64_t row0,
+                      hipStream_t s);
 void launch_transpose_add_lower(double *A, int64_t n, hipStream_t s);
 void launch_tril(const double *F, double *L, int64_t n, hipStream_t s);
 void launch_triu(const double *F, double *U, int64_t n, hipStream_t s);

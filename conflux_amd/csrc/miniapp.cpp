@@ -5,31 +5,24 @@
 //                   [-l print_limit] [-t type] [--sim] [--timing]
 //
 // Differences from the reference (documented in DESIGN.md):
-//   * multi-rank runs self-spawn one process per GPU (fork + pipes for the
-//     RCCL unique id) instead of requiring mpirun;
+//   * multi-rank runs self-spawn one process per GPU (fork+exec, shared
+//     topology in selfspawn.hpp) instead of requiring mpirun;
+//   * when --p_grid is not given, (Px,Py,Pz) is picked from the process
+//     count exactly like the reference (lu_params.hpp:21-47 get_p_grid):
+//     P = CONFLUX_WORLD if set, else the number of visible GPUs;
 //   * --sim runs all ranks of the grid in ONE process on ONE GPU
 //     (choreography-identical, D2D transport) — used for 1-GPU validation;
 //   * --timing disables factor collection (the reference's non-VALIDATION
 //     build); default keeps it on like CONFLUX_WITH_VALIDATION.
-#include <hip/hip_runtime.h>
-#include <sys/wait.h>
-#include <unistd.h>
-
-#include <cmath>
-#include <cstdio>
-#include <cstdlib>
-#include <cstring>
-#include <string>
-#include <vector>
-
-#include "../../include/conflux_lu.h"
+#include "selfspawn.hpp"
 
 static void usage() {
     std::printf(
         "conflux miniapp (MI355X engine)\n"
         "  -N, --cols N          matrix dimension (default 1000 -> rounded)\n"
         "  -b, --block_size b    tile size v (default 256)\n"
-        "  -p, --p_grid Px,Py,Pz process grid (default 1,1,1)\n"
+        "  -p, --p_grid Px,Py,Pz process grid (default: from GPU count,\n"
+        "                        reference lu_params.hpp:21-47 heuristic)\n"
         "  -r, --n_rep r         repetitions (default 2)\n"
         "  -l, --print_limit l   (accepted for compatibility)\n"
         "  -t, --type t          weak|strong|other (label only)\n"
@@ -38,7 +31,7 @@ static void usage() {
 }
 
 int main(int argc, char **argv) {
-    int N = 1000, b = 256, reps = 2, Px = 1, Py = 1, Pz = 1;
+    int N = 1000, b = 256, reps = 2, Px = 0, Py = 0, Pz = 0;
     std::string type = "other";
     bool sim = false, timing = false;
     for (int i = 1; i < argc; ++i) {
@@ -66,7 +59,15 @@ int main(int argc, char **argv) {
             return 1;
         }
     }
-    if (Px <= 0 || Py <= 0 || Pz <= 0) { Px = Py = Pz = 1; }
+    if (Px <= 0 || Py <= 0 || Pz <= 0) {
+        // no grid given: derive it from the process count like the reference
+        // does from the MPI world (lu_params.hpp:21-47)
+        int P = 0;
+        if (const char *w = std::getenv("CONFLUX_WORLD")) P = std::atoi(w);
+        if (P <= 0 && !sim) P = conflux_probe_gpu_count();
+        if (P <= 0) P = 1;
+        conflux_grid_from_P(P, &Px, &Py, &Pz);
+    }
     const int P = Px * Py * Pz;
     // round N like the reference (lu_params.hpp:67-71)
     const int ntx = (N + b * Px - 1) / (b * Px);
@@ -77,68 +78,15 @@ int main(int argc, char **argv) {
     char uid[CONFLUX_LU_UID_BYTES];
     if (sim || P == 1) {
         rank = sim ? -1 : 0;
-        world = P;
     } else if (const char *er = std::getenv("CONFLUX_RANK")) {
         rank = std::atoi(er);
-        // uid passed via fd 3 is only for self-spawn; external launchers use
-        // CONFLUX_UID_FILE
-        if (const char *uf = std::getenv("CONFLUX_UID_FILE")) {
-            FILE *f = std::fopen(uf, "rb");
-            if (!f || std::fread(uid, 1, sizeof uid, f) != sizeof uid) {
-                std::fprintf(stderr, "cannot read uid file\n");
-                return 1;
-            }
-            std::fclose(f);
-        }
+        if (conflux_resolve_uid(uid)) return 1;
     }
-    // (rank == -2 falls through to the self-spawn path below)
+    if (rank == -2)
+        conflux_selfspawn(P, argc, argv,
+                          "--p_grid=" + std::to_string(Px) + "," +
+                              std::to_string(Py) + "," + std::to_string(Pz));
 
-    if (rank == -2) {
-        // self-spawn path
-        int ndev = 0;
-        (void)hipGetDeviceCount(&ndev);
-        if (ndev < P) {
-            std::fprintf(stderr,
-                         "[conflux_miniapp] %d GPUs visible but grid needs %d "
-                         "(use --sim for single-GPU validation)\n",
-                         ndev, P);
-            return 1;
-        }
-        std::vector<int> pipes(2 * P);
-        for (int r = 0; r < P; ++r)
-            if (pipe(&pipes[2 * r])) { perror("pipe"); return 1; }
-        std::vector<pid_t> pids(P);
-        for (int r = 0; r < P; ++r) {
-            pid_t pid = fork();
-            if (pid == 0) {
-                (void)hipSetDevice(r);
-                if (r == 0) {
-                    if (conflux_lu_make_uid(uid)) return 1;
-                    for (int q = 1; q < P; ++q)
-                        if (write(pipes[2 * q + 1], uid, sizeof uid) !=
-                            (ssize_t)sizeof uid)
-                            return 1;
-                } else {
-                    if (read(pipes[2 * r], uid, sizeof uid) !=
-                        (ssize_t)sizeof uid)
-                        return 1;
-                }
-                rank = r;
-                goto run;
-            }
-            pids[r] = pid;
-        }
-        {
-            int status = 0, bad = 0;
-            for (int r = 0; r < P; ++r) {
-                waitpid(pids[r], &status, 0);
-                if (!WIFEXITED(status) || WEXITSTATUS(status)) bad = 1;
-            }
-            return bad;
-        }
-    }
-
-run:
     conflux_lu_ctx *ctx = nullptr;
     int rc = conflux_lu_create(N, b, Px, Py, Pz, rank, world,
                                (world > 1) ? uid : nullptr, &ctx);
@@ -172,9 +120,11 @@ run:
             std::printf("_result_ lu,conflux,%d,%d,%d,%dx%dx%d,time,%s,%.0f,%d\n",
                         N, N_base, P, Px, Py, Pz, type.c_str(), ms, b);
     }
-    if (!timing && (sim || P == 1)) {
+    if (!timing) {
         // reference prints ||PA-LU||_F under CONFLUX_WITH_VALIDATION
-        // (conflux_miniapp.cpp:480-500); here computed on device
+        // (conflux_miniapp.cpp:480-500); here computed on device — in the
+        // distributed case a collective over RCCL (every rank calls;
+        // rank 0 computes and broadcasts)
         double resid = 0;
         if (conflux_lu_validate(ctx, &resid) == 0 && print0)
             std::printf("relative residual ||PA-LU||_F/||A||_F = %.3e\n",
