@@ -90,7 +90,11 @@ static inline int conflux_resolve_uid(char *uid) {
 // status.  Never returns.
 static inline void conflux_selfspawn(int P, int argc, char **argv,
                                      const std::string &extra_arg) {
-    const int ndev = conflux_probe_gpu_count();
+    // CONFLUX_SPAWN_OVERSUBSCRIBE=1: allow more ranks than GPUs, all
+    // sharing device 0 — only meaningful under the shimccl test transport
+    // (real RCCL refuses duplicate GPUs)
+    const bool oversub = std::getenv("CONFLUX_SPAWN_OVERSUBSCRIBE") != nullptr;
+    const int ndev = oversub ? P : conflux_probe_gpu_count();
     if (ndev < P) {
         std::fprintf(stderr,
                      "[conflux] %d GPUs visible but grid needs %d "
@@ -114,7 +118,7 @@ static inline void conflux_selfspawn(int P, int argc, char **argv,
             std::snprintf(wbuf, sizeof wbuf, "%d", P);
             setenv("CONFLUX_RANK", rbuf, 1);
             setenv("CONFLUX_WORLD", wbuf, 1);
-            setenv("HIP_VISIBLE_DEVICES", rbuf, 1);
+            if (!oversub) setenv("HIP_VISIBLE_DEVICES", rbuf, 1);
             setenv("CONFLUX_UID_FILE", uidpath, 1);
             if (r == 0) setenv("CONFLUX_UID_WRITE", "1", 1);
             std::vector<char *> nargv(argv, argv + argc);
