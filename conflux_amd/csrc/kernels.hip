@@ -385,12 +385,17 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                         ppar * PANEL_NB * 8 + 16 * l,
                     0, /*sc1*/ 16);
         }
-        if (tid < 64) drain_stores();
+        // key_abs is payload like the slabs: store it BEFORE the joint
+        // drain so ONE vmcnt covers slabs + abs and the flag store follows
+        // immediately (one L2 round trip per column instead of two on the
+        // last publisher's critical path)
         if (tid == 0) {
             union { double d; unsigned long long u; } a;
             a.d = wa;
             st_rlx_u64(&sync->key_abs[ppar][bid], a.u);
-            drain_stores();
+        }
+        if (tid < 64) drain_stores();
+        if (tid == 0) {
             st_rlx_u64(&sync->key_flag[ppar][bid],
                        ((unsigned long long)pep << 32) | (unsigned)wrow);
             if (pc >= bid * PANEL_RPB && pc < (bid + 1) * PANEL_RPB)

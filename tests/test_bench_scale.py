@@ -1,0 +1,36 @@
+"""Bench-scale correctness: the sizes the bench lines are quoted on
+self-check every round (correctness evidence previously
+stopped at N=4096 while the bench ran N>=16384 timing-only).
+
+One store_factors factorization per bench size through the miniapp CLI
+(factor collection on, like the reference's CONFLUX_WITH_VALIDATION build),
+asserting the device-side stripe-streamed ||PA-LU||_F/||A||_F residual.
+"""
+import os
+import re
+import subprocess
+
+import pytest
+
+REPO = os.path.join(os.path.dirname(os.path.abspath(__file__)), "..")
+MINIAPP = os.path.join(REPO, "conflux_amd", "conflux_miniapp")
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.mark.parametrize("N,v,timeout", [
+    (16384, 512, 900),    # BASELINE cfg 2 (the 1-GPU bench line)
+    (65536, 512, 1800),   # BASELINE cfg 4 size on one GPU (README large-N
+                          # datapoint; exercises the >2^31-element launches)
+])
+def test_single_gpu_validated(N, v, timeout):
+    if not os.path.exists(MINIAPP):
+        pytest.skip("conflux_miniapp not built")
+    out = subprocess.run(
+        [MINIAPP, "-N", str(N), "-b", str(v), "--p_grid=1,1,1", "-r", "1"],
+        capture_output=True, text=True, timeout=timeout)
+    assert out.returncode == 0, out.stdout + out.stderr
+    m = re.search(r"relative residual \|\|PA-LU\|\|_F/\|\|A\|\|_F = (\S+)",
+                  out.stdout)
+    assert m, out.stdout
+    assert float(m.group(1)) < 1e-13

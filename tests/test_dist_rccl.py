@@ -119,7 +119,8 @@ def test_miniapp_selfspawn_dist(tmp_path):
         [MINIAPP, "-N", "1024", "-b", "128", "--p_grid=2,2,1", "-r", "1"],
         env=env, capture_output=True, text=True, timeout=600)
     assert out.returncode == 0, out.stdout + out.stderr
-    res = [l for l in out.stdout.splitlines() if l.startswith("_result_ lu")]
+    res = [l for l in out.stdout.splitlines()
+           if l.startswith("_result_ lu,conflux,1")]  # data line, not header
     assert len(res) == 1, out.stdout
     parts = res[0].split(",")
     assert parts[2] == "1024" and parts[4] == "4" and parts[5] == "2x2x1"
