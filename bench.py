@@ -43,9 +43,12 @@ FP64_MFMA_PEAK_TFLOPS = 78.6  # gfx950: 256 CU x 2.4 GHz x 128 f64 flop/clk/CU
 
 def cpu_baseline():
     """Time the compiled reference (oracle/_ref, MKL+MPICH — kind
-    'reference') on this box's host cores.  Bounded sample: N=8192, v=512,
-    grid 2x2x1 under mpiexec -n 4 (the reference cannot run 1-wide grids),
-    one repetition (~10-30 s of CPU work).  Reported baseline, not target."""
+    'reference') on this box's host cores AT THE BENCH CONFIG's N (16384,
+    v=512 — same size the GPU number beside it is quoted on; the matrix is
+    2 GiB and fits host RAM).  Grid 4x4x1 under mpiexec -n 16 (the
+    reference cannot run 1-wide grids; best measured rank/thread layout on
+    this host class, tools/cpu_sweep.sh), one repetition (~20-30 s of CPU
+    work).  Reported baseline, not target."""
     ref = os.path.join(os.path.dirname(os.path.abspath(__file__)), "oracle",
                        "_ref", "conflux_ref")
     if not os.path.exists(ref):
@@ -60,7 +63,7 @@ def cpu_baseline():
         grid = ("2", "2", "1")
     env = dict(os.environ, MKL_THREADING_LAYER="GNU",
                OMP_NUM_THREADS=str(omp), LD_LIBRARY_PATH="/opt/conda/lib")
-    N, v = 8192, 512
+    N, v = 16384, 512
     try:
         out = subprocess.run(
             ["/opt/conda/bin/mpiexec", "-n", str(n_ranks), ref, str(N),

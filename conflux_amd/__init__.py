@@ -160,14 +160,15 @@ class Engine:
 
     def validate(self):
         """Device-side ||PA - LU||_F / ||A||_F of the last factorization
-        (conflux_lu_validate; sim / single-process only)."""
+        (conflux_lu_validate; stripe-streamed).  COLLECTIVE when world > 1:
+        every rank must call; all ranks return the broadcast residual."""
         r = ctypes.c_double()
         _chk(lib().conflux_lu_validate(self._h, ctypes.byref(r)), "validate")
         return r.value
 
     def validate_cholesky(self):
         """Device-side ||A - L L^T||_F / ||A||_F of the last Cholesky
-        factorization (sim / single-process only)."""
+        factorization.  COLLECTIVE when world > 1 (like validate)."""
         r = ctypes.c_double()
         _chk(lib().conflux_chol_validate(self._h, ctypes.byref(r)),
              "chol_validate")

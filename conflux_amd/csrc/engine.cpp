@@ -1147,7 +1147,8 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         static int env_cap = -1;
         if (env_cap < 0) {
             const char *e = getenv("CONFLUX_GEMM_CAP");
-            env_cap = e ? atoi(e) : 432;
+            env_cap = e ? atoi(e) : 400;  // r02 sweep: 400 beats 432/384/352
+                                          // at N=16384 (218 vs 222 ms/step)
         }
         // Only ranks that RUN step k+1's panel factor need the cap (and
         // the residency guard); ranks outside column k+1 overlap only the
@@ -1580,7 +1581,10 @@ int conflux_lu_create(int N, int v, int Px, int Py, int Pz, int rank,
                       int world, const char *nccl_uid, conflux_lu_ctx **out) {
     if (N <= 0 || v <= 0 || Px <= 0 || Py <= 0 || Pz <= 0) return CONFLUX_LU_EARG;
     if (Px != Py || (Px & (Px - 1)) || v % Pz != 0) return CONFLUX_LU_EARG;
-    if (N % (v * Px) != 0) return CONFLUX_LU_EARG;
+    // round N up to a multiple of v*Px exactly like the reference
+    // (lu_params.hpp:67-71); conflux_lu_dims reports the padded size, and
+    // the generator fills the padding like the reference's InitMatrix does
+    N = v * Px * ((N + v * Px - 1) / (v * Px));
     const int P = Px * Py * Pz;
     const bool sim = (rank < 0);
     if (!sim && world != P) return CONFLUX_LU_EARG;

@@ -677,10 +677,38 @@ __global__ __launch_bounds__(256) void k_trsm_right_upper32(
 #define GEMM_BK 16
 #define GEMM_TPB 256
 
+// tile order: strips of `strip_w` tile-columns walked row-major, so the
+// ~512 concurrently-resident workgroups form a 2D window (e.g. 64 wg =
+// 8 rows x 8 cols at strip_w 8) instead of a 1D row slice.  Cuts the
+// B-panel re-read traffic from ntm full passes to one pass per strip that
+// stays L2/LLC-hot, and keeps each A row-panel hot for strip_w consecutive
+// tiles.  Bijective for ragged tails; strip_w <= 1 is the flat order.
+DEVFN void gemm_tile_of(int wg, int ntm, int ntn, int strip_w, int &tm,
+                        int &tn) {
+    if (strip_w > 1 && strip_w < ntn) {
+        const int per = ntm * strip_w;
+        const int full = ntn / strip_w;
+        const int strip = wg / per;
+        if (strip < full) {
+            const int rem = wg % per;
+            tm = rem / strip_w;
+            tn = strip * strip_w + rem % strip_w;
+        } else {
+            const int rem = wg - full * per;
+            const int wc = ntn - full * strip_w;
+            tm = rem / wc;
+            tn = full * strip_w + rem % wc;
+        }
+    } else {
+        tm = wg / ntn;
+        tn = wg % ntn;
+    }
+}
+
 __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
     const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
     int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
-    int ntm, int ntn) {
+    int ntm, int ntn, int strip_w) {
     // bijective XCD swizzle (guide §5: q/r form)
     int wg = blockIdx.x;
     {
@@ -691,7 +719,8 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
         wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
         if (nwg < 8) wg = blockIdx.x;   // tiny grids: identity
     }
-    const int tm = wg / ntn, tn = wg % ntn;
+    int tm, tn;
+    gemm_tile_of(wg, ntm, ntn, strip_w, tm, tn);
     const int row0 = tm * GEMM_BM;
     const int64_t col0 = (int64_t)tn * GEMM_BN;
 
@@ -806,7 +835,7 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
 __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
     const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
     int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
-    int ntm, int ntn) {
+    int ntm, int ntn, int strip_w) {
     __shared__ double As[2][GEMM_BK][GEMM_BM + 1];   // transposed, padded
     __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];   // double-buffered
     const int nwg = ntm * ntn;
@@ -824,7 +853,8 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
         wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
         if (nwg < 8) wg = vwg;          // tiny grids: identity
     }
-    const int tm = wg / ntn, tn = wg % ntn;
+    int tm, tn;
+    gemm_tile_of(wg, ntm, ntn, strip_w, tm, tn);
     const int row0 = tm * GEMM_BM;
     const int64_t col0 = (int64_t)tn * GEMM_BN;
 
@@ -943,7 +973,7 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
 __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8_nt(
     const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
     int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
-    int ntm, int ntn) {
+    int ntm, int ntn, int strip_w) {
     // bijective XCD swizzle (guide §5: q/r form)
     int wg = blockIdx.x;
     {
@@ -954,7 +984,8 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8_nt(
         wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
         if (nwg < 8) wg = blockIdx.x;   // tiny grids: identity
     }
-    const int tm = wg / ntn, tn = wg % ntn;
+    int tm, tn;
+    gemm_tile_of(wg, ntm, ntn, strip_w, tm, tn);
     const int row0 = tm * GEMM_BM;
     const int64_t col0 = (int64_t)tn * GEMM_BN;
 
@@ -1079,7 +1110,7 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8_nt(
 __global__ __launch_bounds__(512, 4) void k_dgemm_f64_glds(
     const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
     int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
-    int ntm, int ntn) {
+    int ntm, int ntn, int strip_w) {
     int wg = blockIdx.x;
     {
         const int nwg = ntm * ntn;
@@ -1088,7 +1119,8 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_glds(
         wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
         if (nwg < 8) wg = blockIdx.x;
     }
-    const int tm = wg / ntn, tn = wg % ntn;
+    int tm, tn;
+    gemm_tile_of(wg, ntm, ntn, strip_w, tm, tn);
     const int row0 = tm * GEMM_BM;
     const int64_t col0 = (int64_t)tn * GEMM_BN;
 
@@ -1798,6 +1830,15 @@ void launch_potrf32(double *A, int64_t lda, int nb, hipStream_t s) {
 
 int g_dgemm_variant = -1;  // 0 = 4-wave, 1 = 8-wave; env CONFLUX_GEMM_VARIANT
 
+static int gemm_strip_w() {  // strip width in tiles; 0/1 = flat order
+    static int w = -1;
+    if (w < 0) {
+        const char *e = getenv("CONFLUX_GEMM_STRIP");
+        w = e ? atoi(e) : 8;
+    }
+    return w;
+}
+
 void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
                       int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
                       int K, hipStream_t s, int maxwg) {
@@ -1808,18 +1849,19 @@ void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
     }
     const int ntm = (int)cdiv64(M, GEMM_BM);
     const int ntn = (int)cdiv64(N, GEMM_BN);
+    const int sw = gemm_strip_w();
     int nwg = ntm * ntn;
     // cap (persistent w8 kernel only): leave CUs free for a concurrent panel
     if (maxwg > 0 && g_dgemm_variant == 1 && nwg > maxwg) nwg = maxwg;
     if (g_dgemm_variant == 2)
         hipLaunchKernelGGL(k_dgemm_f64_glds, dim3(ntm * ntn), dim3(512), 0, s,
-                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
+                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn, sw);
     else if (g_dgemm_variant == 1)
         hipLaunchKernelGGL(k_dgemm_f64_w8, dim3(nwg), dim3(512), 0, s,
-                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
+                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn, sw);
     else
         hipLaunchKernelGGL(k_dgemm_f64, dim3(ntm * ntn), dim3(GEMM_TPB), 0, s,
-                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
+                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn, sw);
 }
 
 void launch_dgemm_f64_nt(const double *A, int64_t lda, const double *B,
@@ -1829,7 +1871,7 @@ void launch_dgemm_f64_nt(const double *A, int64_t lda, const double *B,
     const int ntm = (int)cdiv64(M, GEMM_BM);
     const int ntn = (int)cdiv64(N, GEMM_BN);
     hipLaunchKernelGGL(k_dgemm_f64_w8_nt, dim3(ntm * ntn), dim3(512), 0, s, A,
-                       lda, B, ldb, C, ldc, M, N, K, ntm, ntn);
+                       lda, B, ldb, C, ldc, M, N, K, ntm, ntn, gemm_strip_w());
 }
 
 void launch_pack_candidate(const double *A10, int64_t lda, const int *gri,
