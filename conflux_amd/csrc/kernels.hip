@@ -316,7 +316,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     double *__restrict__ panel, int64_t ldp, int m, int nb,
     PanelSync2 *__restrict__ sync, int *__restrict__ ipiv,
     unsigned int epoch0, int nblocks, int *__restrict__ swap_dst,
-    int *__restrict__ swap_src) {
+    int *__restrict__ swap_src, int backoff) {
     const int tid = threadIdx.x, bid = blockIdx.x;
     const int r0 = bid * PANEL_RPB + tid;
     const int r1 = r0 + PANEL_TPB;
@@ -469,6 +469,9 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                 unsigned spins = 0;
                 while (((g = ld_rlx_u64(&sync->key_flag[par][b])) >> 32) !=
                        epoch) {
+                    if (backoff == 1) __builtin_amdgcn_s_sleep(1);
+                    else if (backoff == 2) __builtin_amdgcn_s_sleep(2);
+                    else if (backoff >= 4) __builtin_amdgcn_s_sleep(4);
                     if (++spins > 800000000u) {
                         st_rlx_u32(&sync->err, 1u + (unsigned)c);
                         break;
@@ -835,7 +838,7 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
 __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
     const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
     int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
-    int ntm, int ntn, int strip_w) {
+    int ntm, int ntn, int strip_w, int ntc) {
     __shared__ double As[2][GEMM_BK][GEMM_BM + 1];   // transposed, padded
     __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];   // double-buffered
     const int nwg = ntm * ntn;
@@ -956,7 +959,17 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
                 const int r = row0 + wm0 + i * 16 + q * 4 + fk;
                 const int64_t cidx = col0 + wn0 + j * 16 + frow;
                 if (r < M && cidx < N) {
-                    C[(int64_t)r * ldc + cidx] -= acc[i][j][q];
+                    double *cp = &C[(int64_t)r * ldc + cidx];
+                    if (ntc) {
+                        // non-temporal RMW: C has zero reuse — keep its
+                        // 4.3 GB/launch from evicting hot lines (the
+                        // concurrent panel's sync slabs) out of the LLC
+                        __builtin_nontemporal_store(
+                            __builtin_nontemporal_load(cp) - acc[i][j][q],
+                            cp);
+                    } else {
+                        *cp -= acc[i][j][q];
+                    }
                 }
             }
         }
@@ -1404,12 +1417,17 @@ void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
 int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
                         int *ipiv, unsigned int epoch0, int *swap_dst,
                         int *swap_src, hipStream_t s) {
+    static int backoff = -1;
+    if (backoff < 0) {
+        const char *e = getenv("CONFLUX_PANEL_SLEEP");
+        backoff = e ? atoi(e) : 0;
+    }
     int nblocks = (int)cdiv64(m, PANEL_RPB);
     if (nblocks < 1) nblocks = 1;
     if (nblocks > CONFLUX_PANEL_MAX_BLOCKS) return -1;  // not resident: refuse
     hipLaunchKernelGGL(k_panel_factor, dim3(nblocks), dim3(PANEL_TPB), 0, s,
                        panel, ldp, m, nb, (PanelSync2 *)sync, ipiv, epoch0,
-                       nblocks, swap_dst, swap_src);
+                       nblocks, swap_dst, swap_src, backoff);
     return 0;
 }
 
@@ -1839,6 +1857,15 @@ static int gemm_strip_w() {  // strip width in tiles; 0/1 = flat order
     return w;
 }
 
+static int gemm_ntc() {  // non-temporal C epilogue (w8 kernel)
+    static int v = -1;
+    if (v < 0) {
+        const char *e = getenv("CONFLUX_GEMM_NTC");
+        v = e ? atoi(e) : 0;
+    }
+    return v;
+}
+
 void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
                       int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
                       int K, hipStream_t s, int maxwg) {
@@ -1858,7 +1885,8 @@ void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
                            A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn, sw);
     else if (g_dgemm_variant == 1)
         hipLaunchKernelGGL(k_dgemm_f64_w8, dim3(nwg), dim3(512), 0, s,
-                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn, sw);
+                           A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn, sw,
+                           gemm_ntc());
     else
         hipLaunchKernelGGL(k_dgemm_f64, dim3(ntm * ntn), dim3(GEMM_TPB), 0, s,
                            A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn, sw);
