@@ -1420,7 +1420,8 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
     static int backoff = -1;
     if (backoff < 0) {
         const char *e = getenv("CONFLUX_PANEL_SLEEP");
-        backoff = e ? atoi(e) : 0;
+        backoff = e ? atoi(e) : 1;  // r02 matrix: s_sleep(1) in the key poll
+                                    // measured best-or-equal in context
     }
     int nblocks = (int)cdiv64(m, PANEL_RPB);
     if (nblocks < 1) nblocks = 1;
