@@ -179,19 +179,25 @@ def main():
         # therefore understates the kernel: it sustains 53.6 TF/s standalone
         # on the full chip (68% of spec peak, tools/gemm_bench + DESIGN §5).
         achieved = g["flops"] / g["seconds"] / 1e12
+        # Per-launch HBM bytes at the bench GEMM shape, PMC-measured
+        # (rocprofv3 --pmc FETCH_SIZE / WRITE_SIZE in separate passes,
+        # profiles/r02_pmc_dgemm.csv) and CALIBRATED on a known byte count
+        # as the guide prescribes: WRITE_SIZE*1024 / exact C-write bytes
+        # (N^2*8) = 1.4531 at both measured shapes, so raw (FETCH+WRITE)
+        # KB * 1024 / 1.4531.  Result: ~1.0x the algorithmic A+B+2C bytes
+        # at 16384 and ~1.13x at 49152 (strip-ordered tiles; r01's "3x"
+        # was the misapplied x2 wide-read correction + flat tile order).
+        traffic = {
+            (16384, 512): 4.33e9,
+            (49152, 512): 4.43e10,
+        }.get((N, K))
         roofline = {
             "bound": "mfma",
             "achieved": round(achieved, 3),
             "peak": FP64_MFMA_PEAK_TFLOPS,
             "unit": "TFLOP/s",
             "frac": round(achieved / FP64_MFMA_PEAK_TFLOPS, 4),
-            # HBM traffic per launch at the headline shape (M=N=16384,
-            # K=512), rocprofv3 PMC FETCH_SIZE x2 (gfx950 wide-read
-            # correction) + WRITE_SIZE; profiles/r01_pmc_dgemm_*.csv.
-            # 3x the 4.43 GB algorithmic bytes: the BK=16 tile re-reads of
-            # the A/B panels reach the L2 fabric but are served by the
-            # 256 MB Infinity Cache (both panels fit), not DRAM.
-            "traffic": 13.2e9 if (N == 16384 and K == 512) else None,
+            "traffic": traffic,
         }
 
     eng.close()

@@ -492,6 +492,19 @@ int factor_loop(Ctx &c, double *elapsed_ms) {
         t.launches += 1;
         t.flops += c.evs[i].flops;
     }
+    static int spin_stats = -1;
+    if (spin_stats < 0) {
+        const char *e = getenv("CONFLUX_SPIN_STATS");
+        spin_stats = e ? atoi(e) : 0;
+    }
+    if (spin_stats) {
+        for (auto &r : c.rs) {
+            unsigned long long sp2 = 0;
+            conflux_panel_spin_read(r.sync, &sp2, c.stream);
+            std::fprintf(stderr, "[spinstats] rank %d spins=%llu\n",
+                         r.grank, sp2);
+        }
+    }
     return 0;
 }
 

@@ -236,6 +236,9 @@ struct PanelSync2 {
     double diag_row[2][PANEL_NB];
     unsigned int diag_flag[2];
     unsigned int err;
+    // diagnostics: total key-poll iterations (all blocks, all columns) —
+    // read+reset by conflux_panel_spin_stats (CONFLUX_SPIN_STATS=1)
+    unsigned long long spin_sum;
 };
 
 
@@ -338,6 +341,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     const auto srsrc = __builtin_amdgcn_make_buffer_rsrc(
         (void *)sync, (short)0, (int)sizeof(PanelSync2), 0x00020000);
 
+    unsigned long long spin_acc = 0;  // key-poll iterations (diagnostics)
     for (int q = 0; q < 2; ++q) {
         const int r = q ? r1 : r0;
         for (int cc = 0; cc < nb; ++cc)
@@ -477,6 +481,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                         break;
                     }
                 }
+                spin_acc += spins;
                 union { double d; unsigned long long u; } a;
                 a.u = ld_rlx_u64(&sync->key_abs[par][b]);
                 const int rr = (int)(g & 0xffffffffu);
@@ -604,6 +609,14 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         if (r >= m) continue;
         for (int cc = 0; cc < nb; ++cc)
             panel[(int64_t)r * ldp + cc] = rows[q][tid][cc];
+    }
+    // spin diagnostics: wave-0 lanes aggregated, one atomicAdd per block
+    if (tid < 64) {
+        unsigned long long t = spin_acc;
+        for (int w2 = 32; w2 > 0; w2 >>= 1) t += __shfl_down(t, w2);
+        if (tid == 0 && t)
+            (void)__hip_atomic_fetch_add((gu64 *)&sync->spin_sum, t,
+                                         RLX_AGENT);
     }
     // block 0: emit the composed row-permutation map (identity-padded by
     // duplicating entry 0 — benign identical concurrent writes downstream).
@@ -1430,6 +1443,15 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
                        panel, ldp, m, nb, (PanelSync2 *)sync, ipiv, epoch0,
                        nblocks, swap_dst, swap_src, backoff);
     return 0;
+}
+
+void conflux_panel_spin_read(void *sync, unsigned long long *out,
+                             hipStream_t s) {
+    (void)hipMemcpyAsync(out, (char *)sync + offsetof(PanelSync2, spin_sum),
+                         8, hipMemcpyDeviceToHost, s);
+    (void)hipStreamSynchronize(s);
+    (void)hipMemsetAsync((char *)sync + offsetof(PanelSync2, spin_sum), 0, 8,
+                         s);
 }
 
 int conflux_panel_sync_bytes() { return (int)sizeof(PanelSync2); }

@@ -29,6 +29,8 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
                         int *ipiv, unsigned int epoch0, int *swap_dst,
                         int *swap_src, hipStream_t s);
 int conflux_panel_sync_bytes();
+void conflux_panel_spin_read(void *sync, unsigned long long *out,
+                             hipStream_t s);
 int conflux_panel_nb();
 int conflux_panel_rpb();
 void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
