@@ -102,7 +102,9 @@ def main():
     n_gpus = args.gpus or world
 
     # pin this process to its GPU BEFORE the HIP runtime loads
-    if world > 1:
+    # (CONFLUX_BENCH_SHARE_GPU: all ranks on device 0 — only for the
+    # shimccl rehearsal of this launch path on a 1-GPU box)
+    if world > 1 and not os.environ.get("CONFLUX_BENCH_SHARE_GPU"):
         os.environ["HIP_VISIBLE_DEVICES"] = str(local_rank)
 
     N, v, Px, Py, Pz = GRIDS[n_gpus]

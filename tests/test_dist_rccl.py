@@ -106,6 +106,31 @@ def test_dist_parity_vs_oracle(tmp_path, grid, N, v, reps):
     assert np.abs(F - ref["F"]).max() < 1e-11
 
 
+def test_bench_dist_launch(tmp_path):
+    """Rehearse the EXACT multi-rank bench launch the driver uses at round
+    end (torch.distributed.run, one rank per GPU) on one GPU via shimccl:
+    gloo bootstrap, uid broadcast, distributed engine, barriers, the MAX
+    reduction, and the JSON contract line."""
+    import json
+    if not os.path.exists(SHIM):
+        pytest.skip("shimccl.so not built (make -C tests)")
+    env = _dist_env(tmp_path)
+    env["CONFLUX_BENCH_SHARE_GPU"] = "1"
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29532", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1", "--N", "1024",
+         "--v", "128", "--skip-cpu-baseline"],
+        env=env, capture_output=True, text=True, timeout=600, cwd=REPO)
+    assert out.returncode == 0, out.stdout + out.stderr
+    lines = [l for l in out.stdout.splitlines() if l.startswith('{"metric"')]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2 and d["config"]["grid"] == "1x1x2"
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+
+
 def test_miniapp_selfspawn_dist(tmp_path):
     """The CLI self-spawn launcher end to end: fork+exec per rank, uid file
     handoff, distributed factor + validate, `_result_` contract."""
