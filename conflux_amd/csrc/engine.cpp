@@ -250,7 +250,10 @@ int factor_panel(Ctx &c, RankState &r, int n, std::vector<int> &ipiv_out) {
     ipiv_out.assign(v, 0);
     size_t slot;
     if (ev_begin(c, 1, 0, &slot)) return CONFLUX_LU_EHIP;
-    for (int jb = 0; jb < nsteps; jb += NB) {
+    // one NB-wide leaf: persistent factor kernel + full-width rowperm (the
+    // factor kernel composed the realized sub-panel-relative permutation;
+    // applied in two parallel passes over the non-sub-panel columns)
+    auto leaf = [&](int jb) -> int {
         const int nb = std::min(NB, nsteps - jb);
         const int m = n - jb;  // rows of the sub-panel
         if (launch_panel_factor(r.panel + i64(jb) * v + jb, v, m, nb, r.sync,
@@ -260,23 +263,69 @@ int factor_panel(Ctx &c, RankState &r, int n, std::vector<int> &ipiv_out) {
             return CONFLUX_LU_EINTERNAL;
         }
         c.epoch += nb;
-        // apply the sub-panel's swaps to the rest of the panel width: the
-        // factor kernel composed the realized row permutation (sub-panel-
-        // relative), applied here in two parallel passes over the
-        // non-sub-panel columns
         if (v > nb)
             launch_rowperm_skip(r.panel, v, r.d_swap, r.d_swap + 64, jb,
                                 2 * nb, jb, nb, v - nb, r.rowtmp, c.stream);
-        if (jb + nb < v && m > nb) {
-            // U block: rows jb..jb+nb of cols jb+nb..v
-            launch_trsm_left_lower_unit32(r.panel + i64(jb) * v + jb, v,
-                                          r.panel + i64(jb) * v + jb + nb, v,
-                                          nb, v - jb - nb, c.stream);
-            // trailing sub-panel update
-            launch_dgemm_f64(r.panel + i64(jb + nb) * v + jb, v,
-                             r.panel + i64(jb) * v + jb + nb, v,
-                             r.panel + i64(jb + nb) * v + jb + nb, v, m - nb,
-                             v - jb - nb, nb, c.stream);
+        return 0;
+    };
+    static int rec_env = -1;
+    if (rec_env < 0) {
+        const char *e = getenv("CONFLUX_PANEL_REC");
+        rec_env = e ? atoi(e) : 1;
+    }
+    const int nleaf = nsteps / NB;
+    if (rec_env && nsteps == v && v % NB == 0 && nleaf > 1 &&
+        (nleaf & (nleaf - 1)) == 0) {
+        // recursive right-looking glue (LAPACK xGETRF2 shape): leaves stay
+        // NB-wide, but the inter-leaf updates happen at power-of-two nodes
+        // as ONE rank-h TRSM + GEMM of the right half — the 15 K=32 glue
+        // GEMMs per 512-panel become 4 levels of K=32..256 MFMA passes.
+        // Row swaps stay eager full-width at the leaves, so operand rows
+        // are always fully permuted (laswp-both-sides equivalence).
+        int rc2 = 0;
+        std::function<void(int, int)> rec = [&](int a, int b) {
+            if (rc2) return;
+            if (b - a <= NB) {
+                rc2 = leaf(a);
+                return;
+            }
+            const int h = (b - a) / 2;
+            rec(a, a + h);
+            if (rc2) return;
+            // U block: rows a..a+h of cols a+h..b  (L of the left half)
+            if (h == 32)
+                launch_trsm_left_lower_unit32(r.panel + i64(a) * v + a, v,
+                                              r.panel + i64(a) * v + a + h, v,
+                                              h, b - a - h, c.stream);
+            else
+                launch_trsm_left_mfma(r.panel + i64(a) * v + a, v,
+                                      r.panel + i64(a) * v + a + h, v, h,
+                                      b - a - h, c.stream);
+            if (n > a + h)
+                launch_dgemm_f64(r.panel + i64(a + h) * v + a, v,
+                                 r.panel + i64(a) * v + a + h, v,
+                                 r.panel + i64(a + h) * v + a + h, v,
+                                 n - (a + h), b - a - h, h, c.stream);
+            rec(a + h, b);
+        };
+        rec(0, v);
+        if (rc2) return rc2;
+    } else {
+        for (int jb = 0; jb < nsteps; jb += NB) {
+            const int nb = std::min(NB, nsteps - jb);
+            const int m = n - jb;
+            if (leaf(jb)) return CONFLUX_LU_EINTERNAL;
+            if (jb + nb < v && m > nb) {
+                // U block: rows jb..jb+nb of cols jb+nb..v
+                launch_trsm_left_lower_unit32(r.panel + i64(jb) * v + jb, v,
+                                              r.panel + i64(jb) * v + jb + nb,
+                                              v, nb, v - jb - nb, c.stream);
+                // trailing sub-panel update
+                launch_dgemm_f64(r.panel + i64(jb + nb) * v + jb, v,
+                                 r.panel + i64(jb) * v + jb + nb, v,
+                                 r.panel + i64(jb + nb) * v + jb + nb, v,
+                                 m - nb, v - jb - nb, nb, c.stream);
+            }
         }
     }
     if (ev_end(c, slot)) return CONFLUX_LU_EHIP;
