@@ -268,10 +268,16 @@ int factor_panel(Ctx &c, RankState &r, int n, std::vector<int> &ipiv_out) {
                                 2 * nb, jb, nb, v - nb, r.rowtmp, c.stream);
         return 0;
     };
+    // MEASURED ABLATION (r02, default OFF): the recursion is numerically
+    // fine (all parity tests green) but SLOWER — 232.5 vs 223.5 ms/step
+    // in context, 150 vs 133 ms panel sequential at N=16384.  The fat-K
+    // node updates don't pay: node GEMMs are narrow (N = 32..256 cols →
+    // quarter-chip grids) where the flat chain's K=32 GEMMs span the full
+    // remaining width, and the h>=64 node TRSMs cost ~100-500 us each.
     static int rec_env = -1;
     if (rec_env < 0) {
         const char *e = getenv("CONFLUX_PANEL_REC");
-        rec_env = e ? atoi(e) : 1;
+        rec_env = e ? atoi(e) : 0;
     }
     const int nleaf = nsteps / NB;
     if (rec_env && nsteps == v && v % NB == 0 && nleaf > 1 &&
