@@ -84,6 +84,8 @@ def _assemble_F(results, N, v, Px, Py, Pz):
     ((2, 2, 1), 1024, 128, 2),   # butterfly, A00 exchange, spreads; 2 reps
     ((2, 2, 2), 1024, 128, 1),   # the full 3D choreography (BASELINE cfg-4
                                  # grid at test scale)
+    ((2, 2, 1), 4096, 512, 1),   # the BASELINE tile size: v=512 panels,
+                                 # tournament across 2 rank rows
 ])
 def test_dist_parity_vs_oracle(tmp_path, grid, N, v, reps):
     from oracle import Params, gen_matrix, lu_oracle
@@ -129,6 +131,28 @@ def test_bench_dist_launch(tmp_path):
     d = json.loads(lines[0])
     assert d["n_gpus"] == 2 and d["config"]["grid"] == "1x1x2"
     assert d["value"] > 0 and d["ms_per_step"] > 0
+
+
+@pytest.mark.parametrize("grid", ["2,2,1", "2,2,2"])
+def test_cholesky_dist(tmp_path, grid):
+    """The CONFCHOX distributed branches (depth reduce, L_kk column
+    broadcast, slab + transpose spreads) executed multi-process through
+    shimccl, with the distributed Cholesky validation."""
+    chol = os.path.join(REPO, "conflux_amd", "cholesky_miniapp")
+    if not os.path.exists(SHIM):
+        pytest.skip("shimccl.so not built (make -C tests)")
+    if not os.path.exists(chol):
+        pytest.skip("cholesky_miniapp not built")
+    env = _dist_env(tmp_path)
+    env["CONFLUX_SPAWN_OVERSUBSCRIBE"] = "1"
+    out = subprocess.run(
+        [chol, "--dim", "2048", "--tile", "256", "--grid", grid,
+         "--run", "1"],
+        env=env, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stdout + out.stderr
+    resid = [l for l in out.stdout.splitlines() if "relative residual" in l]
+    assert resid, out.stdout
+    assert float(resid[0].split("=")[-1]) < 1e-13
 
 
 def test_miniapp_selfspawn_dist(tmp_path):
