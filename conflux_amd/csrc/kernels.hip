@@ -315,15 +315,18 @@ __global__ __launch_bounds__(256) void k_trsm_left_lower_unit32(
 // the slab handshake; the row swap is performed by the OWNING blocks from
 // slab-published rows (no cross-block matrix reads, no write races).
 // ---------------------------------------------------------------------------
+template <int QR>  // rows per thread: QR*TPB rows/block (QR=2 default;
+                   // QR=1 halves LDS so 2 blocks/CU and halves the
+                   // per-column local update)
 __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     double *__restrict__ panel, int64_t ldp, int m, int nb,
     PanelSync2 *__restrict__ sync, int *__restrict__ ipiv,
     unsigned int epoch0, int nblocks, int *__restrict__ swap_dst,
     int *__restrict__ swap_src, int backoff) {
+    constexpr int RPB = QR * PANEL_TPB;
     const int tid = threadIdx.x, bid = blockIdx.x;
-    const int r0 = bid * PANEL_RPB + tid;
-    const int r1 = r0 + PANEL_TPB;
-    __shared__ double rows[2][PANEL_TPB][PANEL_NB + 1];
+    const int r0 = bid * RPB + tid;
+    __shared__ double rows[QR][PANEL_TPB][PANEL_NB + 1];
     __shared__ double piv_lds[PANEL_NB];
     __shared__ double diag_lds[PANEL_NB];
     __shared__ double red_abs[PANEL_TPB];
@@ -342,8 +345,8 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         (void *)sync, (short)0, (int)sizeof(PanelSync2), 0x00020000);
 
     unsigned long long spin_acc = 0;  // key-poll iterations (diagnostics)
-    for (int q = 0; q < 2; ++q) {
-        const int r = q ? r1 : r0;
+    for (int q = 0; q < QR; ++q) {
+        const int r = r0 + q * PANEL_TPB;
         for (int cc = 0; cc < nb; ++cc)
             rows[q][tid][cc] = (r < m) ? panel[(int64_t)r * ldp + cc] : 0.0;
     }
@@ -359,7 +362,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         const unsigned int pep = epoch0 + (unsigned)pc;
         const int ppar = (int)(pep & 1u);
         if (tid < 16) {
-            const int lrow = (wrow < m) ? wrow - bid * PANEL_RPB : 0;
+            const int lrow = (wrow < m) ? wrow - bid * RPB : 0;
             const int wq = lrow >= PANEL_TPB;
             const int wtid = lrow - wq * PANEL_TPB;
             F64x2Bits x;
@@ -373,10 +376,10 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                             8 +
                         16 * tid,
                     0, /*sc1*/ 16);
-        } else if (tid < 32 && pc >= bid * PANEL_RPB &&
-                   pc < (bid + 1) * PANEL_RPB) {
+        } else if (tid < 32 && pc >= bid * RPB &&
+                   pc < (bid + 1) * RPB) {
             const int l = tid - 16;
-            const int lrow = pc - bid * PANEL_RPB;
+            const int lrow = pc - bid * RPB;
             const int dq = lrow >= PANEL_TPB;
             const int dtid = lrow - dq * PANEL_TPB;
             F64x2Bits x;
@@ -402,7 +405,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         if (tid == 0) {
             st_rlx_u64(&sync->key_flag[ppar][bid],
                        ((unsigned long long)pep << 32) | (unsigned)wrow);
-            if (pc >= bid * PANEL_RPB && pc < (bid + 1) * PANEL_RPB)
+            if (pc >= bid * RPB && pc < (bid + 1) * RPB)
                 st_rlx_u32(&sync->diag_flag[ppar], pep);
         }
     };
@@ -440,8 +443,8 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     {
         double amax = -1.0;
         int arow = m;
-        for (int q = 0; q < 2; ++q) {
-            const int r = q ? r1 : r0;
+        for (int q = 0; q < QR; ++q) {
+            const int r = r0 + q * PANEL_TPB;
             if (r >= 0 && r < m) {
                 const double a = fabs(rows[q][tid][0]);
                 if (a > amax || (a == amax && r < arow)) { amax = a; arow = r; }
@@ -562,8 +565,8 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         // ---- update own rows; track the next column's candidate inline --
         double namax = -1.0;
         int narow = m;
-        for (int q = 0; q < 2; ++q) {
-            const int r = q ? r1 : r0;
+        for (int q = 0; q < QR; ++q) {
+            const int r = r0 + q * PANEL_TPB;
             if (r >= m) continue;
             double *my = rows[q][tid];
             if (r == c) {
@@ -604,8 +607,8 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     }
 
     // write back (plain stores; next kernels see them at the launch boundary)
-    for (int q = 0; q < 2; ++q) {
-        const int r = q ? r1 : r0;
+    for (int q = 0; q < QR; ++q) {
+        const int r = r0 + q * PANEL_TPB;
         if (r >= m) continue;
         for (int cc = 0; cc < nb; ++cc)
             panel[(int64_t)r * ldp + cc] = rows[q][tid][cc];
@@ -1427,6 +1430,15 @@ void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
                        skipn, tot_cols);
 }
 
+static int panel_qr() {  // rows-per-thread variant: CONFLUX_PANEL_RPB
+    static int qr = -1;
+    if (qr < 0) {
+        const char *e = getenv("CONFLUX_PANEL_RPB");
+        qr = (e && atoi(e) == 256) ? 1 : 2;
+    }
+    return qr;
+}
+
 int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
                         int *ipiv, unsigned int epoch0, int *swap_dst,
                         int *swap_src, hipStream_t s) {
@@ -1436,12 +1448,19 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
         backoff = e ? atoi(e) : 1;  // r02 matrix: s_sleep(1) in the key poll
                                     // measured best-or-equal in context
     }
-    int nblocks = (int)cdiv64(m, PANEL_RPB);
+    const int qr = panel_qr();
+    const int rpb = qr * PANEL_TPB;
+    int nblocks = (int)cdiv64(m, rpb);
     if (nblocks < 1) nblocks = 1;
     if (nblocks > CONFLUX_PANEL_MAX_BLOCKS) return -1;  // not resident: refuse
-    hipLaunchKernelGGL(k_panel_factor, dim3(nblocks), dim3(PANEL_TPB), 0, s,
-                       panel, ldp, m, nb, (PanelSync2 *)sync, ipiv, epoch0,
-                       nblocks, swap_dst, swap_src, backoff);
+    if (qr == 1)
+        hipLaunchKernelGGL(k_panel_factor<1>, dim3(nblocks), dim3(PANEL_TPB),
+                           0, s, panel, ldp, m, nb, (PanelSync2 *)sync, ipiv,
+                           epoch0, nblocks, swap_dst, swap_src, backoff);
+    else
+        hipLaunchKernelGGL(k_panel_factor<2>, dim3(nblocks), dim3(PANEL_TPB),
+                           0, s, panel, ldp, m, nb, (PanelSync2 *)sync, ipiv,
+                           epoch0, nblocks, swap_dst, swap_src, backoff);
     return 0;
 }
 
@@ -1456,7 +1475,7 @@ void conflux_panel_spin_read(void *sync, unsigned long long *out,
 
 int conflux_panel_sync_bytes() { return (int)sizeof(PanelSync2); }
 int conflux_panel_nb() { return PANEL_NB; }
-int conflux_panel_rpb() { return PANEL_RPB; }
+int conflux_panel_rpb() { return panel_qr() * PANEL_TPB; }
 
 void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
                                    int64_t ldx, int nb, int64_t N,
