@@ -92,7 +92,8 @@ def main():
     ap.add_argument("--steps", type=int, default=2)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--N", type=int, default=None, help="override matrix dim")
-    ap.add_argument("--v", type=int, default=None, help="override tile size")
+    ap.add_argument("--tile", "--v", dest="v", type=int, default=None,
+                    help="override tile size")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -176,10 +177,11 @@ def main():
     roofline = None
     if g["launches"] > 0 and g["seconds"] > 0:
         # NOTE: with the 1-GPU lookahead overlap the trailing-GEMM launches
-        # in the timed region run on a deliberately CAPPED grid (432/512
+        # in the timed region run on a deliberately CAPPED grid (400/512
         # workgroup slots so the panel kernel keeps whole CUs) — `achieved`
-        # therefore understates the kernel: it sustains 53.6 TF/s standalone
-        # on the full chip (68% of spec peak, tools/gemm_bench + DESIGN §5).
+        # therefore understates the kernel: it sustains 54.2 TF/s standalone
+        # at this shape and 57.5-57.7 TF/s (73% of spec peak) at the
+        # N=49152/65536 north-star shapes (tools/gemm_bench + DESIGN §5).
         achieved = g["flops"] / g["seconds"] / 1e12
         # Per-launch HBM bytes at the bench GEMM shape, PMC-measured
         # (rocprofv3 --pmc FETCH_SIZE / WRITE_SIZE in separate passes,

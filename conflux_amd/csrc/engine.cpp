@@ -1350,6 +1350,9 @@ int trsm_right_lowT(Ctx &c, RankState &r, double *X, int64_t ldx, int M) {
 }
 
 int chol_step(Ctx &c, int k) {
+    c.acomm = c.comm;  // reduce_over_pk sends on acomm; without this the
+                       // Pz>1 depth reduce ran on a NULL communicator
+                       // (caught by the shimccl 2x2x2 distributed test)
     const int v = c.v, Px = c.Px, Py = c.Py, Pz = c.Pz, Nt = c.Nt;
     const int64_t Nl = c.Nl;
     const int kcol = k % Py, krow = k % Px;

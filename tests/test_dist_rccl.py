@@ -79,15 +79,16 @@ def _assemble_F(results, N, v, Px, Py, Pz):
     return F
 
 
-@pytest.mark.parametrize("grid,N,v,reps", [
-    ((1, 1, 2), 1024, 128, 1),   # depth reduce + gpivots broadcast
-    ((2, 2, 1), 1024, 128, 2),   # butterfly, A00 exchange, spreads; 2 reps
-    ((2, 2, 2), 1024, 128, 1),   # the full 3D choreography (BASELINE cfg-4
-                                 # grid at test scale)
-    ((2, 2, 1), 4096, 512, 1),   # the BASELINE tile size: v=512 panels,
-                                 # tournament across 2 rank rows
+@pytest.mark.parametrize("grid,N,v,reps,ftol", [
+    ((1, 1, 2), 1024, 128, 1, 1e-11),  # depth reduce + gpivots broadcast
+    ((2, 2, 1), 1024, 128, 2, 1e-11),  # butterfly, A00 exchange, spreads
+    ((2, 2, 2), 1024, 128, 1, 1e-11),  # the full 3D choreography
+    ((2, 2, 1), 4096, 512, 1, 1e-9),   # BASELINE tile size: v=512 panels,
+                                       # tournament across 2 rank rows
+                                       # (larger N -> looser element tol;
+                                       # pivots stay bit-exact)
 ])
-def test_dist_parity_vs_oracle(tmp_path, grid, N, v, reps):
+def test_dist_parity_vs_oracle(tmp_path, grid, N, v, reps, ftol):
     from oracle import Params, gen_matrix, lu_oracle
 
     Px, Py, Pz = grid
@@ -105,7 +106,7 @@ def test_dist_parity_vs_oracle(tmp_path, grid, N, v, reps):
     assert max(resids) < 1e-13
     # factors match the oracle
     F = _assemble_F(results, N, v, Px, Py, Pz)
-    assert np.abs(F - ref["F"]).max() < 1e-11
+    assert np.abs(F - ref["F"]).max() < ftol
 
 
 def test_bench_dist_launch(tmp_path):
@@ -123,7 +124,7 @@ def test_bench_dist_launch(tmp_path):
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
          "--master-port", "29532", os.path.join(REPO, "bench.py"),
          "--gpus", "2", "--steps", "2", "--warmup", "1", "--N", "1024",
-         "--v", "128", "--skip-cpu-baseline"],
+         "--tile", "128", "--skip-cpu-baseline"],
         env=env, capture_output=True, text=True, timeout=600, cwd=REPO)
     assert out.returncode == 0, out.stdout + out.stderr
     lines = [l for l in out.stdout.splitlines() if l.startswith('{"metric"')]
