@@ -1233,8 +1233,9 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
             // sequential order and the full-width GEMM instead.
             const int nblocks = (panel_rows + conflux_panel_rpb() - 1) /
                                 conflux_panel_rpb();
+            const int bpc = conflux_panel_blocks_per_cu();
             const int free_cus = 256 - (gcap + 1) / 2;
-            if (nblocks > free_cus) {
+            if ((nblocks + bpc - 1) / bpc > free_cus) {
                 async_look = false;
                 gcap = 0;
             }

@@ -1433,10 +1433,19 @@ void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
 static int panel_qr() {  // rows-per-thread variant: CONFLUX_PANEL_RPB
     static int qr = -1;
     if (qr < 0) {
-        const char *e = getenv("CONFLUX_PANEL_RPB");
-        qr = (e && atoi(e) == 256) ? 1 : 2;
+        const char *e = getenv("CONFLUX_PANEL_SLEEP");  // keep init order tame
+        (void)e;
+        const char *r = getenv("CONFLUX_PANEL_RPB");
+        // r02 A/B at N=16384: 256 rows/block (1 row/thread, 67 KB LDS ->
+        // 2 blocks/CU) beats 512: 205 vs 223 ms/step in context, panel
+        // 122 vs 137 ms sequential.  Default 256.
+        qr = (r && atoi(r) == 512) ? 2 : 1;
     }
     return qr;
+}
+
+int conflux_panel_blocks_per_cu() {  // LDS: 135 KB (QR=2) vs 67 KB (QR=1)
+    return panel_qr() == 1 ? 2 : 1;
 }
 
 int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,

@@ -33,6 +33,7 @@ void conflux_panel_spin_read(void *sync, unsigned long long *out,
                              hipStream_t s);
 int conflux_panel_nb();
 int conflux_panel_rpb();
+int conflux_panel_blocks_per_cu();
 void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
                                    int64_t ldx, int nb, int64_t N,
                                    hipStream_t s);
