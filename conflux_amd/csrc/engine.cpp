@@ -1215,8 +1215,9 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         static int env_cap = -1;
         if (env_cap < 0) {
             const char *e = getenv("CONFLUX_GEMM_CAP");
-            env_cap = e ? atoi(e) : 400;  // r02 sweep: 400 beats 432/384/352
-                                          // at N=16384 (218 vs 222 ms/step)
+            env_cap = e ? atoi(e) : 432;  // r02 re-sweep with 256-row panel
+                                          // blocks: 432 ~= 400 (201.5 ms)
+                                          // > 448 > 464 at N=16384
         }
         // Only ranks that RUN step k+1's panel factor need the cap (and
         // the residency guard); ranks outside column k+1 overlap only the
