@@ -315,22 +315,22 @@ __global__ __launch_bounds__(256) void k_trsm_left_lower_unit32(
 // the slab handshake; the row swap is performed by the OWNING blocks from
 // slab-published rows (no cross-block matrix reads, no write races).
 // ---------------------------------------------------------------------------
-template <int QR>  // rows per thread: QR*TPB rows/block (QR=2 default;
-                   // QR=1 halves LDS so 2 blocks/CU and halves the
-                   // per-column local update)
-__global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
+template <int QR, int TPB>  // rows/block = QR*TPB; QR=1 halves the
+                            // per-column local update (1 row/thread);
+                            // TPB=256 gives 67 KB LDS (2 blocks/CU)
+__global__ __launch_bounds__(TPB) void k_panel_factor(
     double *__restrict__ panel, int64_t ldp, int m, int nb,
     PanelSync2 *__restrict__ sync, int *__restrict__ ipiv,
     unsigned int epoch0, int nblocks, int *__restrict__ swap_dst,
     int *__restrict__ swap_src, int backoff) {
-    constexpr int RPB = QR * PANEL_TPB;
+    constexpr int RPB = QR * TPB;
     const int tid = threadIdx.x, bid = blockIdx.x;
     const int r0 = bid * RPB + tid;
-    __shared__ double rows[QR][PANEL_TPB][PANEL_NB + 1];
+    __shared__ double rows[QR][TPB][PANEL_NB + 1];
     __shared__ double piv_lds[PANEL_NB];
     __shared__ double diag_lds[PANEL_NB];
-    __shared__ double red_abs[PANEL_TPB];
-    __shared__ int red_row[PANEL_TPB];
+    __shared__ double red_abs[TPB];
+    __shared__ int red_row[TPB];
     __shared__ unsigned int sh_info[2];
     // block 0 lane 0 composes the dlaswp row-permutation incrementally as
     // pivots are decided (replaces the separate k_swap_map launch):
@@ -346,7 +346,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
 
     unsigned long long spin_acc = 0;  // key-poll iterations (diagnostics)
     for (int q = 0; q < QR; ++q) {
-        const int r = r0 + q * PANEL_TPB;
+        const int r = r0 + q * TPB;
         for (int cc = 0; cc < nb; ++cc)
             rows[q][tid][cc] = (r < m) ? panel[(int64_t)r * ldp + cc] : 0.0;
     }
@@ -363,8 +363,8 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         const int ppar = (int)(pep & 1u);
         if (tid < 16) {
             const int lrow = (wrow < m) ? wrow - bid * RPB : 0;
-            const int wq = lrow >= PANEL_TPB;
-            const int wtid = lrow - wq * PANEL_TPB;
+            const int wq = lrow >= TPB;
+            const int wtid = lrow - wq * TPB;
             F64x2Bits x;
             x.d[0] = rows[wq][wtid][2 * tid];
             x.d[1] = rows[wq][wtid][2 * tid + 1];
@@ -380,8 +380,8 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
                    pc < (bid + 1) * RPB) {
             const int l = tid - 16;
             const int lrow = pc - bid * RPB;
-            const int dq = lrow >= PANEL_TPB;
-            const int dtid = lrow - dq * PANEL_TPB;
+            const int dq = lrow >= TPB;
+            const int dtid = lrow - dq * TPB;
             F64x2Bits x;
             x.d[0] = rows[dq][dtid][2 * l];
             x.d[1] = rows[dq][dtid][2 * l + 1];
@@ -413,7 +413,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
     auto combine_partials = [&](double &wa, int &wrow) {
         wa = red_abs[0];
         wrow = red_row[0];
-        for (int wv = 1; wv < PANEL_TPB / 64; ++wv) {
+        for (int wv = 1; wv < TPB / 64; ++wv) {
             const double oa = red_abs[wv];
             const int orr = red_row[wv];
             if (oa > wa || (oa == wa && orr < wrow)) { wa = oa; wrow = orr; }
@@ -444,7 +444,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         double amax = -1.0;
         int arow = m;
         for (int q = 0; q < QR; ++q) {
-            const int r = r0 + q * PANEL_TPB;
+            const int r = r0 + q * TPB;
             if (r >= 0 && r < m) {
                 const double a = fabs(rows[q][tid][0]);
                 if (a > amax || (a == amax && r < arow)) { amax = a; arow = r; }
@@ -566,7 +566,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
         double namax = -1.0;
         int narow = m;
         for (int q = 0; q < QR; ++q) {
-            const int r = r0 + q * PANEL_TPB;
+            const int r = r0 + q * TPB;
             if (r >= m) continue;
             double *my = rows[q][tid];
             if (r == c) {
@@ -608,7 +608,7 @@ __global__ __launch_bounds__(PANEL_TPB) void k_panel_factor(
 
     // write back (plain stores; next kernels see them at the launch boundary)
     for (int q = 0; q < QR; ++q) {
-        const int r = r0 + q * PANEL_TPB;
+        const int r = r0 + q * TPB;
         if (r >= m) continue;
         for (int cc = 0; cc < nb; ++cc)
             panel[(int64_t)r * ldp + cc] = rows[q][tid][cc];
@@ -1430,22 +1430,33 @@ void launch_rowperm_skip(double *mat, int64_t ld, const int *dst_idx,
                        skipn, tot_cols);
 }
 
-static int panel_qr() {  // rows-per-thread variant: CONFLUX_PANEL_RPB
-    static int qr = -1;
-    if (qr < 0) {
-        const char *e = getenv("CONFLUX_PANEL_SLEEP");  // keep init order tame
-        (void)e;
+// panel block shape: CONFLUX_PANEL_RPB (rows/block) x CONFLUX_PANEL_TPB
+// (threads).  r02 A/B at N=16384: 256 rows/block 1-row/thread (67 KB LDS,
+// 2 blocks/CU) beats the original 512x256 2-rows/thread: 205 vs 223
+// ms/step in context, panel 122 vs 137 ms sequential.  Default 256x256.
+static void panel_shape(int *qr, int *tpb) {
+    static int g_qr = -1, g_tpb = 0;
+    if (g_qr < 0) {
         const char *r = getenv("CONFLUX_PANEL_RPB");
-        // r02 A/B at N=16384: 256 rows/block (1 row/thread, 67 KB LDS ->
-        // 2 blocks/CU) beats 512: 205 vs 223 ms/step in context, panel
-        // 122 vs 137 ms sequential.  Default 256.
-        qr = (r && atoi(r) == 512) ? 2 : 1;
+        const char *t = getenv("CONFLUX_PANEL_TPB");
+        g_tpb = (t && atoi(t) == 512) ? 512 : 256;
+        const int rpb = r ? atoi(r) : 256;
+        g_qr = (rpb / g_tpb >= 2) ? 2 : 1;
     }
+    *qr = g_qr;
+    *tpb = g_tpb;
+}
+
+static int panel_qr() {
+    int qr, tpb;
+    panel_shape(&qr, &tpb);
     return qr;
 }
 
-int conflux_panel_blocks_per_cu() {  // LDS: 135 KB (QR=2) vs 67 KB (QR=1)
-    return panel_qr() == 1 ? 2 : 1;
+int conflux_panel_blocks_per_cu() {  // LDS/block: QR*TPB*264 B + tails
+    int qr, tpb;
+    panel_shape(&qr, &tpb);
+    return (qr * tpb <= 256) ? 2 : 1;
 }
 
 int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
@@ -1457,17 +1468,22 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
         backoff = e ? atoi(e) : 1;  // r02 matrix: s_sleep(1) in the key poll
                                     // measured best-or-equal in context
     }
-    const int qr = panel_qr();
-    const int rpb = qr * PANEL_TPB;
+    int qr, tpb;
+    panel_shape(&qr, &tpb);
+    const int rpb = qr * tpb;
     int nblocks = (int)cdiv64(m, rpb);
     if (nblocks < 1) nblocks = 1;
     if (nblocks > CONFLUX_PANEL_MAX_BLOCKS) return -1;  // not resident: refuse
-    if (qr == 1)
-        hipLaunchKernelGGL(k_panel_factor<1>, dim3(nblocks), dim3(PANEL_TPB),
+    if (qr == 1 && tpb == 512)
+        hipLaunchKernelGGL((k_panel_factor<1, 512>), dim3(nblocks), dim3(512),
+                           0, s, panel, ldp, m, nb, (PanelSync2 *)sync, ipiv,
+                           epoch0, nblocks, swap_dst, swap_src, backoff);
+    else if (qr == 1)
+        hipLaunchKernelGGL((k_panel_factor<1, 256>), dim3(nblocks), dim3(256),
                            0, s, panel, ldp, m, nb, (PanelSync2 *)sync, ipiv,
                            epoch0, nblocks, swap_dst, swap_src, backoff);
     else
-        hipLaunchKernelGGL(k_panel_factor<2>, dim3(nblocks), dim3(PANEL_TPB),
+        hipLaunchKernelGGL((k_panel_factor<2, 256>), dim3(nblocks), dim3(256),
                            0, s, panel, ldp, m, nb, (PanelSync2 *)sync, ipiv,
                            epoch0, nblocks, swap_dst, swap_src, backoff);
     return 0;
@@ -1484,7 +1500,11 @@ void conflux_panel_spin_read(void *sync, unsigned long long *out,
 
 int conflux_panel_sync_bytes() { return (int)sizeof(PanelSync2); }
 int conflux_panel_nb() { return PANEL_NB; }
-int conflux_panel_rpb() { return panel_qr() * PANEL_TPB; }
+int conflux_panel_rpb() {
+    int qr, tpb;
+    panel_shape(&qr, &tpb);
+    return qr * tpb;
+}
 
 void launch_trsm_left_lower_unit32(const double *L, int64_t ldl, double *X,
                                    int64_t ldx, int nb, int64_t N,
