@@ -1470,9 +1470,14 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
     }
     int qr, tpb;
     panel_shape(&qr, &tpb);
-    const int rpb = qr * tpb;
+    int rpb = qr * tpb;
     int nblocks = (int)cdiv64(m, rpb);
     if (nblocks < 1) nblocks = 1;
+    if (nblocks > CONFLUX_PANEL_MAX_BLOCKS && rpb < 512) {
+        // very tall panel (m > 64k rows): auto-promote to 512-row blocks
+        qr = 2; tpb = 256; rpb = 512;
+        nblocks = (int)cdiv64(m, rpb);
+    }
     if (nblocks > CONFLUX_PANEL_MAX_BLOCKS) return -1;  // not resident: refuse
     if (qr == 1 && tpb == 512)
         hipLaunchKernelGGL((k_panel_factor<1, 512>), dim3(nblocks), dim3(512),
