@@ -1439,7 +1439,8 @@ static void panel_shape(int *qr, int *tpb) {
     if (g_qr < 0) {
         const char *r = getenv("CONFLUX_PANEL_RPB");
         const char *t = getenv("CONFLUX_PANEL_TPB");
-        g_tpb = (t && atoi(t) == 512) ? 512 : 256;
+        g_tpb = t ? atoi(t) : 256;
+        if (g_tpb != 512 && g_tpb != 128) g_tpb = 256;
         const int rpb = r ? atoi(r) : 256;
         g_qr = (rpb / g_tpb >= 2) ? 2 : 1;
     }
@@ -1481,6 +1482,10 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
     if (nblocks > CONFLUX_PANEL_MAX_BLOCKS) return -1;  // not resident: refuse
     if (qr == 1 && tpb == 512)
         hipLaunchKernelGGL((k_panel_factor<1, 512>), dim3(nblocks), dim3(512),
+                           0, s, panel, ldp, m, nb, (PanelSync2 *)sync, ipiv,
+                           epoch0, nblocks, swap_dst, swap_src, backoff);
+    else if (qr == 1 && tpb == 128)
+        hipLaunchKernelGGL((k_panel_factor<1, 128>), dim3(nblocks), dim3(128),
                            0, s, panel, ldp, m, nb, (PanelSync2 *)sync, ipiv,
                            epoch0, nblocks, swap_dst, swap_src, backoff);
     else if (qr == 1)
