@@ -1454,10 +1454,11 @@ static int panel_qr() {
     return qr;
 }
 
-int conflux_panel_blocks_per_cu() {  // LDS/block: QR*TPB*264 B + tails
+int conflux_panel_blocks_per_cu() {  // LDS/block ~= QR*TPB*264 B + tails
     int qr, tpb;
     panel_shape(&qr, &tpb);
-    return (qr * tpb <= 256) ? 2 : 1;
+    const int rows = qr * tpb;
+    return rows <= 128 ? 4 : rows <= 256 ? 2 : 1;  // 34/67/135 KB of 160
 }
 
 int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,

@@ -113,6 +113,7 @@ GRIDS = [
     (128, 16, 2, 2, 2),
     (128, 8, 4, 4, 2),
     (256, 32, 2, 2, 1),
+    (512, 32, 8, 8, 1),   # 3-round butterfly (ceil(log2 8)); Ml = 2v edge
 ]
 
 
