@@ -119,6 +119,7 @@ struct Ctx {
     hipEvent_t ev_pc{};          // panel-columns-updated event
     hipEvent_t ev_t3{};          // step-3 complete (gates the step-5 TRSM)
     hipEvent_t ev_t5{};          // step-5 TRSM complete (gates the C9 spread)
+    hipEvent_t ev_t5a{};         // panel-slice of the A01 solve complete
     std::vector<RankState> rs;   // size P (sim) or 1 (distributed)
     std::vector<int> pivotInds;  // M, global pivot ids (all ranks identical)
     unsigned epoch = 1;
@@ -785,6 +786,9 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
     const int loff = (k / Py) * v;
     const int kcol = k % Py, krow = k % Px;
     const int64_t wA01 = Nl - loff;
+    const bool look = (k + 1 < c.Nt);
+    const int ncol = (k + 1) % Py;
+    const int64_t lnext = i64(v) * ((k + 1) / Py);
     (void)kcol;
 
     // ---- step 2: push pivot rows up, pack, depth-reduce (C7) --------------
@@ -1014,6 +1018,10 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         split_env = se ? atoi(se) : 1;
     }
     const bool split_trsm = !c.sim && c.panel_stream && split_env;
+    // degenerate 1-rank grid: solve the v columns step k+1's panel needs
+    // FIRST and record ev_t5a after them, so the (a) GEMM and the next
+    // panel chain start without waiting for the rest of the A01 solve
+    const bool slice_first = split_trsm && Pz == 1 && Px == 1 && look;
     if (split_trsm) {
         HIPCHK(hipEventRecord(c.ev_t3, c.stream));
         HIPCHK(hipStreamWaitEvent(c.panel_stream, c.ev_t3, 0));
@@ -1021,7 +1029,18 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         c.stream = c.panel_stream;
         for (auto &r : c.rs) {
             if (r.pi != krow || r.pk != 0) continue;
-            if (trsm_left_lower(c, r, r.A01, wA01, wA01))
+            if (slice_first) {
+                const int64_t s0 = lnext - loff;  // slice start in A01
+                if (trsm_left_lower(c, r, r.A01 + s0, wA01, v))
+                    return CONFLUX_LU_EINTERNAL;
+                HIPCHK(hipEventRecord(c.ev_t5a, c.stream));
+                if (s0 > 0 && trsm_left_lower(c, r, r.A01, wA01, s0))
+                    return CONFLUX_LU_EINTERNAL;
+                if (wA01 - s0 - v > 0 &&
+                    trsm_left_lower(c, r, r.A01 + s0 + v, wA01,
+                                    wA01 - s0 - v))
+                    return CONFLUX_LU_EINTERNAL;
+            } else if (trsm_left_lower(c, r, r.A01, wA01, wA01))
                 return CONFLUX_LU_EINTERNAL;
             if (c.store_factors) {
                 const int ltik = k / Px;
@@ -1136,7 +1155,7 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
                                   v, c.stream);
             }
         }
-    } else {
+    } else if (!slice_first) {
         HIPCHK(hipStreamWaitEvent(c.stream, c.ev_t5, 0));
     }
     if (Pz == 1 && Px == 1) {
@@ -1145,7 +1164,14 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
             for (auto &x : c.rs)
                 launch_copy2d(x.A01, wA01, x.A01Rcv, Nl, c.nlayr, wA01,
                               c.stream);
-        else
+        else if (slice_first) {
+            // slice only; the rest is copied after ev_t5 below, before the
+            // (b) trailing pieces that read it
+            HIPCHK(hipStreamWaitEvent(c.stream, c.ev_t5a, 0));
+            const int64_t s0 = lnext - loff;
+            launch_copy2d(r.A01 + s0, wA01, r.A01Rcv + s0, Nl, c.nlayr, v,
+                          c.stream);
+        } else
             launch_copy2d(r.A01, wA01, r.A01Rcv, Nl, c.nlayr, wA01, c.stream);
     } else {
         for (int pj = 0; pj < Py; ++pj) {
@@ -1199,9 +1225,6 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
                          c.nlayr, c.stream, cap);
         return ev_end(c, slot) ? CONFLUX_LU_EHIP : 0;
     };
-    const bool look = (k + 1 < c.Nt);
-    const int ncol = (k + 1) % Py;
-    const int64_t lnext = i64(v) * ((k + 1) / Py);
     const char *lk = getenv("CONFLUX_LOOKAHEAD");
     bool async_look = look && !c.sim && c.panel_stream &&
                       (lk ? atoi(lk) != 0 : true);
@@ -1258,6 +1281,17 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
                         r.A11 + i64(r.fnp) * Nl + lnext, i64(r.nact) * Nl);
     if (look && !c.sim && c.panel_stream) {
         HIPCHK(hipEventRecord(c.ev_pc, c.stream));
+    }
+    if (slice_first) {
+        // rest of the A01 solve feeds the (b) pieces below
+        RankState &r = c.rs[0];
+        HIPCHK(hipStreamWaitEvent(c.stream, c.ev_t5, 0));
+        const int64_t s0 = lnext - loff;
+        if (s0 > 0)
+            launch_copy2d(r.A01, wA01, r.A01Rcv, Nl, c.nlayr, s0, c.stream);
+        if (wA01 - s0 - v > 0)
+            launch_copy2d(r.A01 + s0 + v, wA01, r.A01Rcv + s0 + v, Nl,
+                          c.nlayr, wA01 - s0 - v, c.stream);
     }
     // (b) the rest of the trailing update — capped while the panel runs.
     // It must be ENQUEUED before phase01: phase01 ends in a host sync (the
@@ -1700,7 +1734,8 @@ int conflux_lu_create(int N, int v, int Px, int Py, int Pz, int rank,
         if (hipStreamCreate(&c->panel_stream) != hipSuccess ||
             hipEventCreate(&c->ev_pc) != hipSuccess ||
             hipEventCreate(&c->ev_t3) != hipSuccess ||
-            hipEventCreate(&c->ev_t5) != hipSuccess) {
+            hipEventCreate(&c->ev_t5) != hipSuccess ||
+            hipEventCreate(&c->ev_t5a) != hipSuccess) {
             delete c;
             return CONFLUX_LU_EHIP;
         }
@@ -2189,6 +2224,7 @@ int conflux_lu_destroy(conflux_lu_ctx *c) {
     if (c->ev_pc) (void)hipEventDestroy(c->ev_pc);
     if (c->ev_t3) (void)hipEventDestroy(c->ev_t3);
     if (c->ev_t5) (void)hipEventDestroy(c->ev_t5);
+    if (c->ev_t5a) (void)hipEventDestroy(c->ev_t5a);
     (void)hipStreamDestroy(c->stream);
     delete c;
     return CONFLUX_LU_OK;
