@@ -1020,8 +1020,20 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
     const bool split_trsm = !c.sim && c.panel_stream && split_env;
     // degenerate 1-rank grid: solve the v columns step k+1's panel needs
     // FIRST and record ev_t5a after them, so the (a) GEMM and the next
-    // panel chain start without waiting for the rest of the A01 solve
-    const bool slice_first = split_trsm && Pz == 1 && Px == 1 && look;
+    // panel chain start without waiting for the rest of the A01 solve.
+    // MEASURED ABLATION (r02, default OFF): bit-identical and parity-green
+    // but SLOWER — 229.6 vs 201 ms/step at N=16384.  The fused TRSM
+    // streams all previously-solved 32-wide L panels through LDS once PER
+    // LAUNCH regardless of the X width, so the 3-way column split nearly
+    // triples the solve cost (trsm 33 -> 65 ms/job) and eats the overlap
+    // gain.  Would need a width-proportional TRSM to pay.
+    static int slice_env = -1;
+    if (slice_env < 0) {
+        const char *se2 = getenv("CONFLUX_SLICE_TRSM");
+        slice_env = se2 ? atoi(se2) : 0;
+    }
+    const bool slice_first =
+        split_trsm && slice_env && Pz == 1 && Px == 1 && look;
     if (split_trsm) {
         HIPCHK(hipEventRecord(c.ev_t3, c.stream));
         HIPCHK(hipStreamWaitEvent(c.panel_stream, c.ev_t3, 0));
