@@ -1743,7 +1743,17 @@ int conflux_lu_create(int N, int v, int Px, int Py, int Pz, int rank,
         c->rs.resize(1);
         const int pi = rank / (Py * Pz), pj = (rank / Pz) % Py, pk = rank % Pz;
         if (alloc_rank(*c, c->rs[0], pi, pj, pk)) { delete c; return CONFLUX_LU_EHIP; }
-        if (hipStreamCreate(&c->panel_stream) != hipSuccess ||
+        // the panel chain is the latency-critical path: give its stream
+        // the highest hardware queue priority (CONFLUX_PANEL_PRIO=0 to
+        // disable)
+        const char *pp = getenv("CONFLUX_PANEL_PRIO");
+        int prio_on = pp ? atoi(pp) : 1;
+        int lo = 0, hi = 0;
+        (void)hipDeviceGetStreamPriorityRange(&lo, &hi);
+        if ((prio_on
+                 ? hipStreamCreateWithPriority(&c->panel_stream,
+                                               hipStreamDefault, hi)
+                 : hipStreamCreate(&c->panel_stream)) != hipSuccess ||
             hipEventCreate(&c->ev_pc) != hipSuccess ||
             hipEventCreate(&c->ev_t3) != hipSuccess ||
             hipEventCreate(&c->ev_t5) != hipSuccess ||
