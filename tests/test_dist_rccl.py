@@ -107,6 +107,19 @@ def test_dist_parity_vs_oracle(tmp_path, grid, N, v, reps, ftol):
     # factors match the oracle
     F = _assemble_F(results, N, v, Px, Py, Pz)
     assert np.abs(F - ref["F"]).max() < ftol
+    # ... and are BIT-IDENTICAL to simulation mode: the sim and distributed
+    # paths share every kernel and the deterministic pk-ascending combine
+    # order; only the transport differs, so any divergence is a transport
+    # ordering bug
+    from conflux_amd import Engine
+    with Engine(N, v, Px, Py, Pz, rank=-1) as e:
+        e.store_factors(True)
+        e.init_matrix(42)
+        e.factor()
+        Fsim = e.get_F_global()
+        perm_sim = e.get_perm()
+    assert np.array_equal(perm_sim, results[0]["perm"])
+    assert np.array_equal(Fsim, F), "sim and distributed factors must be bit-identical"
 
 
 def test_bench_dist_launch(tmp_path):
