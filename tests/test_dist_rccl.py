@@ -87,6 +87,9 @@ def _assemble_F(results, N, v, Px, Py, Pz):
                                        # tournament across 2 rank rows
                                        # (larger N -> looser element tol;
                                        # pivots stay bit-exact)
+    ((4, 4, 1), 2048, 128, 1, 1e-10),  # 16 ranks: 2-round butterfly and
+                                       # 4-row pivot routing on the real
+                                       # dist code
 ])
 def test_dist_parity_vs_oracle(tmp_path, grid, N, v, reps, ftol):
     from oracle import Params, gen_matrix, lu_oracle
