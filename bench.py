@@ -191,9 +191,14 @@ def main():
         # KB * 1024 / 1.4531.  Result: ~1.0x the algorithmic A+B+2C bytes
         # at 16384 and ~1.13x at 49152 (strip-ordered tiles; r01's "3x"
         # was the misapplied x2 wide-read correction + flat tile order).
+        # keyed by (global N, nlayr); the value is the measured per-launch
+        # traffic of THAT config's dominant per-rank GEMM shape
         traffic = {
-            (16384, 512): 4.33e9,
-            (49152, 512): 4.43e10,
+            (16384, 512): 4.33e9,   # 1 GPU: launch 16384^2 K=512 (0.98x alg)
+            (49152, 512): 4.43e10,  # 1-GPU cfg-4-scale: 49152^2 K=512 (1.13x)
+            (32768, 512): 4.33e9,   # cfg 3 (4 GPU): per-rank 16384^2 K=512
+            (65536, 256): 1.59e10,  # cfg 4 (8 GPU): per-rank 32768^2 K=256
+                                    # (0.91x algorithmic)
         }.get((N, K))
         roofline = {
             "bound": "mfma",
