@@ -997,6 +997,126 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
 }
 
 
+// 256x128 block-tile variant (CONFLUX_GEMM_VARIANT=3): 8 waves as 4x2 of
+// 64x64 fragments — double the accumulator chains per wave and half the
+// barriers per flop vs the 128x128 w8 kernel, at 99 KB LDS (1 wg/CU,
+// 2 waves/SIMD).  Plain (non-persistent) launch: meant for the uncapped
+// full-chip launches; the capped overlap path keeps the w8 kernel.
+__global__ __launch_bounds__(512) void k_dgemm_f64_bm256(
+    const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
+    int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
+    int ntm, int ntn, int strip_w) {
+    int wg = blockIdx.x;
+    {
+        const int nwg = ntm * ntn;
+        const int q = nwg >> 3, r = nwg & 7;
+        const int xcd = wg & 7, idx = wg >> 3;
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+        if (nwg < 8) wg = blockIdx.x;
+    }
+    int tm, tn;
+    gemm_tile_of(wg, ntm, ntn, strip_w, tm, tn);
+    const int row0 = tm * 256;
+    const int64_t col0 = (int64_t)tn * GEMM_BN;
+
+    __shared__ double As[2][GEMM_BK][256 + 1];   // transposed, padded
+    __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm0 = (wave >> 1) * 64;             // wave's 64x64 sub-tile
+    const int wn0 = (wave & 1) * 64;
+    const int frow = lane & 15;
+    const int fk = lane >> 4;
+
+    f64x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = f64x4{0, 0, 0, 0};
+
+    double ra[8], rb[4];
+    const int ktiles = (K + GEMM_BK - 1) / GEMM_BK;
+
+    auto load_a = [&](int kt) {
+        const int kk = kt * GEMM_BK;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            const int e = tid + i * 512;               // 0..4095
+            const int r = e >> 4, k = e & 15;
+            const int gr = row0 + r;
+            ra[i] = (gr < M && kk + k < K) ? A[(int64_t)gr * lda + kk + k] : 0.0;
+        }
+    };
+    auto load_b = [&](int kt) {
+        const int kk = kt * GEMM_BK;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;               // 0..2047
+            const int k = e >> 7, c = e & 127;
+            const int64_t gc = col0 + c;
+            rb[i] = (kk + k < K && gc < N) ? B[(int64_t)(kk + k) * ldb + gc] : 0.0;
+        }
+    };
+    auto write_lds = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            const int e = tid + i * 512;
+            As[buf][e & 15][e >> 4] = ra[i];
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;
+            Bs[buf][e >> 7][e & 127] = rb[i];
+        }
+    };
+
+    load_a(0);
+    load_b(0);
+    write_lds(0);
+    __syncthreads();
+
+    int cur = 0;
+    for (int kt = 0; kt < ktiles; ++kt) {
+        if (kt + 1 < ktiles) {
+            load_a(kt + 1);
+            load_b(kt + 1);
+        }
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            const int k = kk * 4 + fk;
+            double af[4], bf[4];
+#pragma unroll
+            for (int i = 0; i < 4; ++i) af[i] = As[cur][k][wm0 + i * 16 + frow];
+#pragma unroll
+            for (int j = 0; j < 4; ++j) bf[j] = Bs[cur][k][wn0 + j * 16 + frow];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+        if (kt + 1 < ktiles) write_lds(cur ^ 1);
+        __syncthreads();
+        cur ^= 1;
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int r = row0 + wm0 + i * 16 + q * 4 + fk;
+                const int64_t cidx = col0 + wn0 + j * 16 + frow;
+                if (r < M && cidx < N) C[(int64_t)r * ldc + cidx] -= acc[i][j][q];
+            }
+        }
+    }
+}
+
 // NT variant: C -= A * B^T with B stored (N x K) row-major — the
 // computeA11 low-rank update (reference Cholesky.cpp:345-351).
 __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8_nt(
@@ -1962,6 +2082,12 @@ void launch_dgemm_f64(const double *A, int64_t lda, const double *B,
     int nwg = ntm * ntn;
     // cap (persistent w8 kernel only): leave CUs free for a concurrent panel
     if (maxwg > 0 && g_dgemm_variant == 1 && nwg > maxwg) nwg = maxwg;
+    if (g_dgemm_variant == 3) {
+        const int ntm2 = (int)cdiv64(M, 256);
+        hipLaunchKernelGGL(k_dgemm_f64_bm256, dim3(ntm2 * ntn), dim3(512), 0,
+                           s, A, lda, B, ldb, C, ldc, M, N, K, ntm2, ntn, sw);
+        return;
+    }
     if (g_dgemm_variant == 2)
         hipLaunchKernelGGL(k_dgemm_f64_glds, dim3(ntm * ntn), dim3(512), 0, s,
                            A, lda, B, ldb, C, ldc, M, N, K, ntm, ntn, sw);
