@@ -1019,8 +1019,9 @@ __global__ __launch_bounds__(512) void k_dgemm_f64_bm256(
     const int row0 = tm * 256;
     const int64_t col0 = (int64_t)tn * GEMM_BN;
 
-    __shared__ double As[2][GEMM_BK][256 + 1];   // transposed, padded
-    __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];
+    constexpr int BK8 = 8;  // half-depth K-tile: 49.5 KB LDS -> 2 wg/CU
+    __shared__ double As[2][BK8][256 + 1];   // transposed, padded
+    __shared__ double Bs[2][BK8][GEMM_BN + 2];
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -1036,24 +1037,24 @@ __global__ __launch_bounds__(512) void k_dgemm_f64_bm256(
 #pragma unroll
         for (int j = 0; j < 4; ++j) acc[i][j] = f64x4{0, 0, 0, 0};
 
-    double ra[8], rb[4];
-    const int ktiles = (K + GEMM_BK - 1) / GEMM_BK;
+    double ra[4], rb[2];
+    const int ktiles = (K + BK8 - 1) / BK8;
 
     auto load_a = [&](int kt) {
-        const int kk = kt * GEMM_BK;
+        const int kk = kt * BK8;
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-            const int e = tid + i * 512;               // 0..4095
-            const int r = e >> 4, k = e & 15;
+        for (int i = 0; i < 4; ++i) {
+            const int e = tid + i * 512;               // 0..2047
+            const int r = e >> 3, k = e & 7;
             const int gr = row0 + r;
             ra[i] = (gr < M && kk + k < K) ? A[(int64_t)gr * lda + kk + k] : 0.0;
         }
     };
     auto load_b = [&](int kt) {
-        const int kk = kt * GEMM_BK;
+        const int kk = kt * BK8;
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-            const int e = tid + i * 512;               // 0..2047
+        for (int i = 0; i < 2; ++i) {
+            const int e = tid + i * 512;               // 0..1023
             const int k = e >> 7, c = e & 127;
             const int64_t gc = col0 + c;
             rb[i] = (kk + k < K && gc < N) ? B[(int64_t)(kk + k) * ldb + gc] : 0.0;
@@ -1061,12 +1062,12 @@ __global__ __launch_bounds__(512) void k_dgemm_f64_bm256(
     };
     auto write_lds = [&](int buf) {
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
+        for (int i = 0; i < 4; ++i) {
             const int e = tid + i * 512;
-            As[buf][e & 15][e >> 4] = ra[i];
+            As[buf][e & 7][e >> 3] = ra[i];
         }
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
+        for (int i = 0; i < 2; ++i) {
             const int e = tid + i * 512;
             Bs[buf][e >> 7][e & 127] = rb[i];
         }
@@ -1084,7 +1085,7 @@ __global__ __launch_bounds__(512) void k_dgemm_f64_bm256(
             load_b(kt + 1);
         }
 #pragma unroll
-        for (int kk = 0; kk < 4; ++kk) {
+        for (int kk = 0; kk < 2; ++kk) {
             const int k = kk * 4 + fk;
             double af[4], bf[4];
 #pragma unroll
