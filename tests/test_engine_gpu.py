@@ -400,3 +400,21 @@ def test_miniapp_cli_full_cfg4_single_gpu():
                   out.stdout)
     assert m, out.stdout
     assert int(m.group(1)) < 60000  # sanity: minutes would mean spin stalls
+
+
+@pytest.mark.parametrize("seed", [7, 1234, 987654321])
+def test_lu_parity_seeds(eng, seed):
+    """Seed fuzz: the device generator and the whole pipeline stay
+    pivot-bit-exact vs the oracle on inputs other than the bench seed."""
+    N, v, Px, Py, Pz = 256, 32, 2, 2, 1
+    A = gen_matrix(N, seed)
+    r = lu_oracle(A, Params(N, v, Px, Py, Pz))
+    with eng.Engine(N, v, Px, Py, Pz, rank=-1) as e:
+        e.store_factors(True)
+        e.init_matrix(seed)            # device-side generator, same seed
+        e.factor()
+        perm = e.get_perm()
+        F = e.get_F_global()
+    assert np.array_equal(perm, r["perm"])
+    assert np.abs(F - r["F"]).max() < TOL_F
+    assert residual_check(A, perm, F) < TOL_RES
