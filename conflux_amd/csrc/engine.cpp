@@ -1601,17 +1601,54 @@ int chol_step(Ctx &c, int k) {
     }
 
     // ---- c4: low-rank updates of local tiles with i >= j > k --------------
+    // One rectangular launch per rank with the tile-diagonal mask (v % 128
+    // == 0): workgroups above the global tile diagonal exit immediately, so
+    // the whole trapezoid updates at full-chip GEMM efficiency instead of
+    // one narrow launch per column tile (~970 launches/factorization at
+    // N=16384 averaging 22 TF; profiles/r02_kernel_stats_chol_n16384.csv).
+    // A01Rcv's per-tile v x nlayr slabs are globally contiguous, so it IS
+    // the (Nl x nlayr) B^T operand.  Tiles with global row tile < col tile
+    // are never read anywhere (only the lower triangle is meaningful), so
+    // masked-out regions are simply untouched.
+    static int rect_env = -1;
+    if (rect_env < 0) {
+        const char *e = getenv("CONFLUX_CHOL_RECT");
+        rect_env = e ? atoi(e) : 1;
+    }
     for (auto &r : c.rs) {
         const int f2 = f2_of(r.pi);
-        for (int ltj = 0; ltj < c.tA11y; ++ltj) {
+        const int ltj0 = (r.pj <= k) ? (k - r.pj) / Py + 1 : 0;
+        const int r0 = v * ntiles_lt(c, r.pi, k + 1);
+        double fl = 0;  // algorithmic flops (valid tiles only)
+        for (int ltj = ltj0; ltj < c.tA11y; ++ltj) {
+            const int gtj = ltj * Py + r.pj;
+            if (gtj >= Nt) continue;
+            const int M2 = c.Ml - v * ntiles_lt(c, r.pi, gtj);
+            if (M2 > 0) fl += 2.0 * M2 * (double)v * c.nlayr;
+        }
+        if (rect_env && v % 128 == 0) {
+            const int M2r = c.Ml - r0;
+            const int64_t N2 = Nl - i64(ltj0) * v;
+            if (M2r <= 0 || N2 <= 0 || ltj0 >= c.tA11y) continue;
+            size_t slot;
+            if (ev_begin(c, 0, fl, &slot)) return CONFLUX_LU_EHIP;
+            launch_dgemm_f64_nt_tril(
+                r.A10Rcv + i64(r0 - f2) * c.nlayr, c.nlayr,
+                r.A01Rcv + i64(ltj0) * v * c.nlayr, c.nlayr,
+                r.A11 + i64(r0) * Nl + i64(ltj0) * v, Nl, M2r, N2, c.nlayr,
+                v, r0, i64(ltj0) * v, Px, Py, r.pi, r.pj, c.stream);
+            if (ev_end(c, slot)) return CONFLUX_LU_EHIP;
+            continue;
+        }
+        for (int ltj = ltj0; ltj < c.tA11y; ++ltj) {
             const int gtj = ltj * Py + r.pj;
             if (gtj <= k || gtj >= Nt) continue;
             const int rstart = v * ntiles_lt(c, r.pi, gtj);
             const int M2 = c.Ml - rstart;
             if (M2 <= 0) continue;
-            const double fl = 2.0 * M2 * (double)v * c.nlayr;
+            const double fl1 = 2.0 * M2 * (double)v * c.nlayr;
             size_t slot;
-            if (ev_begin(c, 0, fl, &slot)) return CONFLUX_LU_EHIP;
+            if (ev_begin(c, 0, fl1, &slot)) return CONFLUX_LU_EHIP;
             launch_dgemm_f64_nt(r.A10Rcv + i64(rstart - f2) * c.nlayr, c.nlayr,
                                 r.A01Rcv + i64(ltj) * v * c.nlayr, c.nlayr,
                                 r.A11 + i64(rstart) * Nl + i64(ltj) * v, Nl,

@@ -1120,10 +1120,22 @@ __global__ __launch_bounds__(512) void k_dgemm_f64_bm256(
 
 // NT variant: C -= A * B^T with B stored (N x K) row-major — the
 // computeA11 low-rank update (reference Cholesky.cpp:345-351).
+// Tile-cyclic lower-triangle mask (the Cholesky c4 single-launch update):
+// workgroups whose GLOBAL v-tile row < v-tile col exit immediately, so one
+// rectangular launch covers the trapezoid of tiles with i >= j and spends
+// nothing on the rest.  mask.v == 0 disables.  A 128-wide workgroup never
+// straddles a v-tile boundary (v % 128 == 0 enforced by the launcher).
+struct TriMask {
+    int v;          // global tile size (0 = no mask)
+    int r0off;      // launch-local row 0's offset within the rank-local A11
+    int64_t c0off;  // same for columns
+    int Px, Py, pi, pj;
+};
+
 __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8_nt(
     const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
     int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
-    int ntm, int ntn, int strip_w) {
+    int ntm, int ntn, int strip_w, TriMask mask) {
     // bijective XCD swizzle (guide §5: q/r form)
     int wg = blockIdx.x;
     {
@@ -1138,6 +1150,11 @@ __global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8_nt(
     gemm_tile_of(wg, ntm, ntn, strip_w, tm, tn);
     const int row0 = tm * GEMM_BM;
     const int64_t col0 = (int64_t)tn * GEMM_BN;
+    if (mask.v > 0) {
+        const int gti = (int)((mask.r0off + row0) / mask.v) * mask.Px + mask.pi;
+        const int gtj = (int)((mask.c0off + col0) / mask.v) * mask.Py + mask.pj;
+        if (gti < gtj) return;  // above the global tile diagonal: untouched
+    }
 
     __shared__ double As[2][GEMM_BK][GEMM_BM + 1];   // transposed, padded
     __shared__ double Bs[2][GEMM_BK][GEMM_BN + 2];   // double-buffered
@@ -2108,7 +2125,24 @@ void launch_dgemm_f64_nt(const double *A, int64_t lda, const double *B,
     const int ntm = (int)cdiv64(M, GEMM_BM);
     const int ntn = (int)cdiv64(N, GEMM_BN);
     hipLaunchKernelGGL(k_dgemm_f64_w8_nt, dim3(ntm * ntn), dim3(512), 0, s, A,
-                       lda, B, ldb, C, ldc, M, N, K, ntm, ntn, gemm_strip_w());
+                       lda, B, ldb, C, ldc, M, N, K, ntm, ntn, gemm_strip_w(),
+                       TriMask{0, 0, 0, 0, 0, 0, 0});
+}
+
+// Cholesky c4 single-launch variant: the trapezoid of v-tiles with global
+// tile row >= tile col, launched as one rectangle with the TriMask.
+// Requires v % 128 == 0 (caller falls back to per-tile launches otherwise).
+void launch_dgemm_f64_nt_tril(const double *A, int64_t lda, const double *B,
+                              int64_t ldb, double *C, int64_t ldc, int M,
+                              int64_t N, int K, int v, int r0off,
+                              int64_t c0off, int Px, int Py, int pi, int pj,
+                              hipStream_t s) {
+    if (M <= 0 || N <= 0 || K <= 0) return;
+    const int ntm = (int)cdiv64(M, GEMM_BM);
+    const int ntn = (int)cdiv64(N, GEMM_BN);
+    hipLaunchKernelGGL(k_dgemm_f64_w8_nt, dim3(ntm * ntn), dim3(512), 0, s, A,
+                       lda, B, ldb, C, ldc, M, N, K, ntm, ntn, gemm_strip_w(),
+                       TriMask{v, r0off, c0off, Px, Py, pi, pj});
 }
 
 void launch_pack_candidate(const double *A10, int64_t lda, const int *gri,

@@ -59,6 +59,11 @@ void launch_potrf32(double *A, int64_t lda, int nb, hipStream_t s);
 void launch_dgemm_f64_nt(const double *A, int64_t lda, const double *B,
                          int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
                          int K, hipStream_t s);
+void launch_dgemm_f64_nt_tril(const double *A, int64_t lda, const double *B,
+                              int64_t ldb, double *C, int64_t ldc, int M,
+                              int64_t N, int K, int v, int r0off,
+                              int64_t c0off, int Px, int Py, int pi, int pj,
+                              hipStream_t s);
 void launch_init_matrix_spd(double *A, int Ml, int Nl, int v, int Px, int Py,
                             int pi, int pj, int zero_layer, uint64_t seed,
                             int64_t Nglob, hipStream_t s);
