@@ -1626,7 +1626,13 @@ int chol_step(Ctx &c, int k) {
             const int M2 = c.Ml - v * ntiles_lt(c, r.pi, gtj);
             if (M2 > 0) fl += 2.0 * M2 * (double)v * c.nlayr;
         }
-        if (rect_env && v % 128 == 0) {
+        // adaptive: per-tile launches already fill the chip when a tile
+        // column's grid has >= ~512 workgroups (measured: per-tile wins at
+        // Ml-r0 = 32768, rect wins at <= 16384); batch with the masked
+        // rectangle only when they would underfill
+        const bool use_rect =
+            rect_env && v % 128 == 0 && i64(c.Ml - r0) * 4 / 128 <= 512;
+        if (use_rect) {
             const int M2r = c.Ml - r0;
             const int64_t N2 = Nl - i64(ltj0) * v;
             if (M2r <= 0 || N2 <= 0 || ltj0 >= c.tA11y) continue;
