@@ -67,6 +67,8 @@ struct RankState {
     double *A01 = nullptr;      // v x Nl
     double *A10Rcv = nullptr;   // Ml x nlayr
     double *A01Rcv = nullptr;   // nlayr x Nl
+    double *A10Rcv2 = nullptr;  // second set: the single-rank Cholesky
+    double *A01Rcv2 = nullptr;  // lookahead double-buffers the slabs
     double *A00 = nullptr;      // v x v (packed LU of the pivot block)
     double *cand = nullptr;     // 2v x (v+1)
     double *panel = nullptr;    // max(2v, Ml) x v — getrf workspace
@@ -175,6 +177,7 @@ int ensure_factor_bufs(Ctx &c, RankState &r, bool need_hist = true) {
 
 void free_rank(RankState &r) {
     for (double *p : {r.A11, r.A11in, r.A10, r.A01, r.A10Rcv, r.A01Rcv,
+                      r.A10Rcv2, r.A01Rcv2,
                       r.A00, r.cand, r.panel, r.cm, r.A01pack, r.rowtmp,
                       r.redtmp, r.slabs, r.Fres, r.A10hist})
         if (p) (void)hipFree(p);
@@ -1665,6 +1668,94 @@ int chol_step(Ctx &c, int k) {
     return 0;
 }
 
+// ---------------------------------------------------------------------------
+// single-rank (1x1x1, world == 1) Cholesky lookahead: while step k's
+// trailing update (b) runs on the main stream, step k+1's whole serial
+// chain — c0 column copy, blocked potrf, the rank-v TRSM, the slab copies —
+// runs on the second stream into the OTHER receive-slab set.  Bit-identical
+// to the sequential order: the (a)/(b) column split partitions the same
+// per-tile updates, and the chain is gated on exactly the columns it reads.
+// ---------------------------------------------------------------------------
+int chol_chain_1r(Ctx &c, int k, double *rcvA, double *rcvB) {
+    const int v = c.v;
+    const int64_t Nl = c.Nl;
+    RankState &r = c.rs[0];
+    const int fnp = k * v, f2 = fnp + v;
+    launch_copy2d(r.A11 + i64(fnp) * Nl + i64(k) * v, Nl,
+                  r.A10 + i64(fnp) * v, v, c.Ml - fnp, v, c.stream);
+    if (potrf_tile(c, r.A10 + i64(fnp) * v, v)) return CONFLUX_LU_EINTERNAL;
+    launch_copy2d(r.A10 + i64(fnp) * v, v, r.A00, v, v, v, c.stream);
+    const int n2 = c.Ml - f2;
+    if (n2 > 0 && trsm_right_lowT(c, r, r.A10 + i64(f2) * v, v, n2))
+        return CONFLUX_LU_EINTERNAL;
+    if (c.store_factors) {
+        if (n2 > 0)
+            launch_copy2d(r.A10 + i64(f2) * v, v,
+                          r.Fres + i64(f2) * Nl + i64(k) * v, Nl, n2, v,
+                          c.stream);
+        launch_copy2d(r.A00, v, r.Fres + i64(fnp) * Nl + i64(k) * v, Nl, v,
+                      v, c.stream);
+    }
+    if (n2 > 0) {
+        launch_copy2d(r.A10 + i64(f2) * v, v, rcvA, c.nlayr, n2, v, c.stream);
+        // A01Rcv tiles k+1..Nt are rcvA shifted one tile (1x1x1: nlayr == v)
+        launch_copy2d(rcvA, c.nlayr, rcvB + i64(k + 1) * v * c.nlayr,
+                      c.nlayr, n2, c.nlayr, c.stream);
+    }
+    return 0;
+}
+
+int chol_c4_1r(Ctx &c, int k, double *rcvA, double *rcvB, bool slice_only,
+               bool rest_only) {
+    const int v = c.v;
+    const int64_t Nl = c.Nl;
+    const int Nt = c.Nt;
+    RankState &r = c.rs[0];
+    const int f2 = (k + 1) * v;
+    if (!rest_only && k + 1 < Nt) {  // (a): tile column k+1 only
+        const int M2 = c.Ml - f2;
+        if (M2 > 0) {
+            const double fl = 2.0 * M2 * (double)v * c.nlayr;
+            size_t slot;
+            if (ev_begin(c, 0, fl, &slot)) return CONFLUX_LU_EHIP;
+            launch_dgemm_f64_nt(rcvA, c.nlayr,
+                                rcvB + i64(k + 1) * v * c.nlayr, c.nlayr,
+                                r.A11 + i64(f2) * Nl + f2, Nl, M2, v,
+                                c.nlayr, c.stream);
+            if (ev_end(c, slot)) return CONFLUX_LU_EHIP;
+        }
+    }
+    if (slice_only || k + 2 >= Nt) return 0;
+    // (b): tile columns k+2..Nt (adaptive rect as in chol_step)
+    const int r0 = (k + 2) * v;
+    const int64_t c0 = i64(k + 2) * v;
+    const int M2r = c.Ml - r0;
+    const int64_t N2 = Nl - c0;
+    if (M2r <= 0 || N2 <= 0) return 0;
+    double fl = 0;
+    for (int j = k + 2; j < Nt; ++j) fl += 2.0 * (c.Ml - i64(j) * v) * v * c.nlayr;
+    const bool use_rect = v % 128 == 0 && i64(M2r) * 4 / 128 <= 512;
+    size_t slot;
+    if (ev_begin(c, 0, fl, &slot)) return CONFLUX_LU_EHIP;
+    if (use_rect) {
+        launch_dgemm_f64_nt_tril(rcvA + i64(r0 - f2) * c.nlayr, c.nlayr,
+                                 rcvB + c0 * c.nlayr, c.nlayr,
+                                 r.A11 + i64(r0) * Nl + c0, Nl, M2r, N2,
+                                 c.nlayr, v, r0, c0, 1, 1, 0, 0, c.stream);
+    } else {
+        for (int j = k + 2; j < Nt; ++j) {
+            const int rs = j * v, M2 = c.Ml - rs;
+            if (M2 <= 0) continue;
+            launch_dgemm_f64_nt(rcvA + i64(rs - f2) * c.nlayr, c.nlayr,
+                                rcvB + i64(j) * v * c.nlayr, c.nlayr,
+                                r.A11 + i64(rs) * Nl + i64(j) * v, Nl, M2, v,
+                                c.nlayr, c.stream);
+        }
+    }
+    if (ev_end(c, slot)) return CONFLUX_LU_EHIP;
+    return 0;
+}
+
 int chol_loop(Ctx &c, double *elapsed_ms) {
     for (auto &r : c.rs) {
         if (c.store_factors) {
@@ -1688,9 +1779,44 @@ int chol_loop(Ctx &c, double *elapsed_ms) {
         HIPCHK(hipStreamSynchronize(c.stream));
     }
     const auto t1 = std::chrono::high_resolution_clock::now();
-    for (int k = 0; k < c.Nt; ++k) {
-        int rc = chol_step(c, k);
+    static int chol_look = -1;
+    if (chol_look < 0) {
+        const char *e = getenv("CONFLUX_CHOL_LOOK");
+        chol_look = e ? atoi(e) : 1;
+    }
+    if (chol_look && !c.sim && c.world == 1 && c.Px == 1 && c.Py == 1 &&
+        c.Pz == 1 && c.panel_stream) {
+        RankState &r = c.rs[0];
+        if (!r.A10Rcv2) {
+            HIPCHK(hipMalloc(&r.A10Rcv2, i64(c.Ml) * c.nlayr * 8));
+            HIPCHK(hipMalloc(&r.A01Rcv2, i64(c.nlayr) * c.Nl * 8));
+        }
+        double *As2[2] = {r.A10Rcv, r.A10Rcv2};
+        double *Bs2[2] = {r.A01Rcv, r.A01Rcv2};
+        int rc = chol_chain_1r(c, 0, As2[0], Bs2[0]);
         if (rc) return rc;
+        for (int k = 0; k + 1 < c.Nt; ++k) {
+            const int par = k & 1;
+            if ((rc = chol_c4_1r(c, k, As2[par], Bs2[par], true, false)))
+                return rc;
+            HIPCHK(hipEventRecord(c.ev_pc, c.stream));
+            HIPCHK(hipStreamWaitEvent(c.panel_stream, c.ev_pc, 0));
+            hipStream_t saved = c.stream;
+            c.stream = c.panel_stream;
+            rc = chol_chain_1r(c, k + 1, As2[par ^ 1], Bs2[par ^ 1]);
+            if (!rc && hipEventRecord(c.ev_t5, c.stream) != hipSuccess)
+                rc = CONFLUX_LU_EHIP;
+            c.stream = saved;
+            if (rc) return rc;
+            if ((rc = chol_c4_1r(c, k, As2[par], Bs2[par], false, true)))
+                return rc;
+            HIPCHK(hipStreamWaitEvent(c.stream, c.ev_t5, 0));
+        }
+    } else {
+        for (int k = 0; k < c.Nt; ++k) {
+            int rc = chol_step(c, k);
+            if (rc) return rc;
+        }
     }
     HIPCHK(hipStreamSynchronize(c.stream));
     if (c.have_comm) {

@@ -34,3 +34,20 @@ def test_single_gpu_validated(N, v, timeout):
                   out.stdout)
     assert m, out.stdout
     assert float(m.group(1)) < 1e-13
+
+
+def test_cholesky_single_rank_validated():
+    """The non-sim single-rank Cholesky path (lookahead chain on the second
+    stream + adaptive masked c4 rectangle) validates on device."""
+    chol = os.path.join(REPO, "conflux_amd", "cholesky_miniapp")
+    if not os.path.exists(chol):
+        pytest.skip("cholesky_miniapp not built")
+    out = subprocess.run(
+        [chol, "--dim", "4096", "--tile", "512", "--grid", "1,1,1",
+         "--run", "1"],
+        capture_output=True, text=True, timeout=900)
+    assert out.returncode == 0, out.stdout + out.stderr
+    m = re.search(r"relative residual \|\|A-LL\^T\|\|_F/\|\|A\|\|_F = (\S+)",
+                  out.stdout)
+    assert m, out.stdout
+    assert float(m.group(1)) < 1e-13
