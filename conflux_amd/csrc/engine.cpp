@@ -2351,6 +2351,35 @@ int conflux_lu_debug_dgemm_bench(int M, int64_t N, int K, int iters,
     return CONFLUX_LU_OK;
 }
 
+int conflux_lu_debug_dgemm_nt_bench(int M, int64_t N, int K, int iters,
+                                    double *tflops) {
+    hipStream_t s;
+    HIPCHK(hipStreamCreate(&s));
+    double *dA, *dB, *dC;
+    HIPCHK(hipMalloc(&dA, i64(M) * K * 8));
+    HIPCHK(hipMalloc(&dB, N * K * 8));
+    HIPCHK(hipMalloc(&dC, i64(M) * N * 8));
+    launch_init_matrix(dA, M, K, K, 1, 1, 0, 0, 0, 7, s);
+    launch_init_matrix(dB, (int)N, K, K, 1, 1, 0, 0, 0, 8, s);
+    launch_init_matrix(dC, M, (int)N, (int)N, 1, 1, 0, 0, 0, 9, s);
+    launch_dgemm_f64_nt(dA, K, dB, K, dC, N, M, N, K, s);  // warmup
+    hipEvent_t a, b;
+    HIPCHK(hipEventCreate(&a));
+    HIPCHK(hipEventCreate(&b));
+    HIPCHK(hipEventRecord(a, s));
+    for (int i = 0; i < iters; ++i)
+        launch_dgemm_f64_nt(dA, K, dB, K, dC, N, M, N, K, s);
+    HIPCHK(hipEventRecord(b, s));
+    HIPCHK(hipEventSynchronize(b));
+    float ms = 0;
+    HIPCHK(hipEventElapsedTime(&ms, a, b));
+    *tflops = 2.0 * M * (double)N * K * iters / (ms * 1e-3) / 1e12;
+    (void)hipFree(dA); (void)hipFree(dB); (void)hipFree(dC);
+    (void)hipEventDestroy(a); (void)hipEventDestroy(b);
+    (void)hipStreamDestroy(s);
+    return CONFLUX_LU_OK;
+}
+
 int conflux_lu_debug_getrf(int n, int v, double *panel, int *ipiv_out) {
     Ctx c;
     c.v = v;
