@@ -1735,6 +1735,14 @@ int chol_c4_1r(Ctx &c, int k, double *rcvA, double *rcvB, bool slice_only,
     double fl = 0;
     for (int j = k + 2; j < Nt; ++j) fl += 2.0 * (c.Ml - i64(j) * v) * v * c.nlayr;
     const bool use_rect = v % 128 == 0 && i64(M2r) * 4 / 128 <= 512;
+    static int dbg = -1;
+    if (dbg < 0) {
+        const char *e = getenv("CONFLUX_CHOL_DEBUG");
+        dbg = e ? atoi(e) : 0;
+    }
+    if (dbg)
+        std::fprintf(stderr, "[c4_1r] k=%d M2r=%d rect=%d\n", k, M2r,
+                     (int)use_rect);
     size_t slot;
     if (ev_begin(c, 0, fl, &slot)) return CONFLUX_LU_EHIP;
     if (use_rect) {
