@@ -418,3 +418,21 @@ def test_lu_parity_seeds(eng, seed):
     assert np.array_equal(perm, r["perm"])
     assert np.abs(F - r["F"]).max() < TOL_F
     assert residual_check(A, perm, F) < TOL_RES
+
+
+def test_lu_parity_8192_tournament(eng):
+    """Bit-exact pivot ceiling at the bench tile size: N=8192, v=512,
+    2x2x1 — tournament pivoting over 2 rank rows at real panel widths
+    (the largest size the oracle restatement runs in about a minute)."""
+    N, v = 8192, 512
+    A = gen_matrix(N)
+    r = lu_oracle(A, Params(N, v, 2, 2, 1))
+    with eng.Engine(N, v, 2, 2, 1, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor()
+        perm = e.get_perm()
+        F = e.get_F_global()
+    assert np.array_equal(perm, r["perm"]), "pivots bit-exact at N=8192"
+    assert np.abs(F - r["F"]).max() < 1e-9
+    assert residual_check(A, perm, F) < 1e-13
