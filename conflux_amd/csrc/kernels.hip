@@ -851,7 +851,12 @@ __global__ __launch_bounds__(GEMM_TPB) void k_dgemm_f64(
     }
 }
 
-__global__ __launch_bounds__(512, 2) void k_dgemm_f64_w8(
+// NOTE (r02 measured): __launch_bounds__(512, 4) is load-bearing — it
+// forces a 128-register allocation (with a small scratch spill) whose
+// schedule sustains 57.7 TF; relaxing to (512, 2) (166 regs, no spill,
+// same 2-waves/SIMD occupancy) drops to 44.3, and software-pipelining the
+// fragment loads on top of it reaches only 45.  Keep the tight bound.
+__global__ __launch_bounds__(512, 4) void k_dgemm_f64_w8(
     const double *__restrict__ A, int64_t lda, const double *__restrict__ B,
     int64_t ldb, double *__restrict__ C, int64_t ldc, int M, int64_t N, int K,
     int ntm, int ntn, int strip_w, int ntc) {
@@ -941,29 +946,20 @@ __global__ __launch_bounds__(512, 2) void k_dgemm_f64_w8(
             load_a(kt + 1);
             load_b(kt + 1);
         }
-        double af[2][4], bf[2][2];
-        auto ldfrag = [&](int kk, int p) {
-#pragma unroll
-            for (int i = 0; i < 4; ++i)
-                af[p][i] = As[cur][kk * 4 + fk][wm0 + i * 16 + frow];
-#pragma unroll
-            for (int j = 0; j < 2; ++j)
-                bf[p][j] = Bs[cur][kk * 4 + fk][wn0 + j * 16 + frow];
-        };
-        ldfrag(0, 0);
 #pragma unroll
         for (int kk = 0; kk < 4; ++kk) {
-            // next fragment's LDS reads issue BEFORE this one's MFMAs
-            // (separate registers -> no WAR -> the latency hides under
-            // the 8-MFMA pack instead of a full lgkmcnt(0) stall)
-            if (kk + 1 < 4) ldfrag(kk + 1, (kk + 1) & 1);
-            const int p = kk & 1;
+            const int k = kk * 4 + fk;
+            double af[4], bf[2];
+#pragma unroll
+            for (int i = 0; i < 4; ++i) af[i] = As[cur][k][wm0 + i * 16 + frow];
+#pragma unroll
+            for (int j = 0; j < 2; ++j) bf[j] = Bs[cur][k][wn0 + j * 16 + frow];
 #pragma unroll
             for (int i = 0; i < 4; ++i)
 #pragma unroll
                 for (int j = 0; j < 2; ++j)
                     acc[i][j] = __builtin_amdgcn_mfma_f64_16x16x4f64(
-                        af[p][i], bf[p][j], acc[i][j], 0, 0, 0);
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
         }
         // write t+1 into the other buffer (everyone finished reading it at
         // the barrier that ended step t-1), then one barrier per K-step
