@@ -118,6 +118,10 @@ struct Ctx {
     bool have_pcomm = false;
     hipStream_t stream{};
     hipStream_t panel_stream{};  // lookahead stream (distributed mode)
+    hipStream_t trsm_stream{};   // step-5 A01 solve (r02: its own stream —
+                                 // on panel_stream it serialized against
+                                 // the next panel chain, the saturated
+                                 // pipeline: 175 of 213 ms busy)
     hipEvent_t ev_pc{};          // panel-columns-updated event
     hipEvent_t ev_t3{};          // step-3 complete (gates the step-5 TRSM)
     hipEvent_t ev_t5{};          // step-5 TRSM complete (gates the C9 spread)
@@ -1020,7 +1024,7 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         const char *se = getenv("CONFLUX_SPLIT_TRSM");
         split_env = se ? atoi(se) : 1;
     }
-    const bool split_trsm = !c.sim && c.panel_stream && split_env;
+    const bool split_trsm = !c.sim && c.trsm_stream && split_env;
     // degenerate 1-rank grid: solve the v columns step k+1's panel needs
     // FIRST and record ev_t5a after them, so the (a) GEMM and the next
     // panel chain start without waiting for the rest of the A01 solve.
@@ -1039,9 +1043,9 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
         split_trsm && slice_env && Pz == 1 && Px == 1 && look;
     if (split_trsm) {
         HIPCHK(hipEventRecord(c.ev_t3, c.stream));
-        HIPCHK(hipStreamWaitEvent(c.panel_stream, c.ev_t3, 0));
+        HIPCHK(hipStreamWaitEvent(c.trsm_stream, c.ev_t3, 0));
         hipStream_t saved = c.stream;
-        c.stream = c.panel_stream;
+        c.stream = c.trsm_stream;
         for (auto &r : c.rs) {
             if (r.pi != krow || r.pk != 0) continue;
             if (slice_first) {
@@ -1930,6 +1934,7 @@ int conflux_lu_create(int N, int v, int Px, int Py, int Pz, int rank,
                  ? hipStreamCreateWithPriority(&c->panel_stream,
                                                hipStreamDefault, hi)
                  : hipStreamCreate(&c->panel_stream)) != hipSuccess ||
+            hipStreamCreate(&c->trsm_stream) != hipSuccess ||
             hipEventCreate(&c->ev_pc) != hipSuccess ||
             hipEventCreate(&c->ev_t3) != hipSuccess ||
             hipEventCreate(&c->ev_t5) != hipSuccess ||
@@ -2448,6 +2453,7 @@ int conflux_lu_destroy(conflux_lu_ctx *c) {
     if (c->have_pcomm) (void)ncclCommDestroy(c->pcomm);
     if (c->have_comm) (void)ncclCommDestroy(c->comm);
     if (c->panel_stream) (void)hipStreamDestroy(c->panel_stream);
+    if (c->trsm_stream) (void)hipStreamDestroy(c->trsm_stream);
     if (c->ev_pc) (void)hipEventDestroy(c->ev_pc);
     if (c->ev_t3) (void)hipEventDestroy(c->ev_t3);
     if (c->ev_t5) (void)hipEventDestroy(c->ev_t5);
