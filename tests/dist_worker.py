@@ -8,7 +8,11 @@ standing in for RCCL (all ranks share one GPU).  With SHIMCCL_DIR fixed,
 ncclGetUniqueId is deterministic, so every rank derives the same uid locally
 and no out-of-band exchange is needed.
 
-usage: dist_worker.py N v Px Py Pz rank reps out.npz
+usage: dist_worker.py N v Px Py Pz rank reps out.npz [set_matrix]
+
+With the optional 9th arg "set_matrix", the rank uploads its tile-cyclic
+local slice through conflux_lu_set_matrix_local (the documented drop-in
+data path) instead of the device-side generator — same matrix either way.
 """
 import os
 import sys
@@ -26,11 +30,25 @@ def main():
 
     P = Px * Py * Pz
     uid = Engine.make_uid()
+    use_set = len(sys.argv) > 9 and sys.argv[9] == "set_matrix"
     with Engine(N, v, Px, Py, Pz, rank=rank, world=P, uid=uid) as e:
         e.store_factors(True)
         ms = 0.0
         for _ in range(reps):
-            e.init_matrix(42)
+            if use_set:
+                from oracle import gen_matrix
+                pi, pj, pk = rank // (Py * Pz), (rank // Pz) % Py, rank % Pz
+                loc = np.zeros((e.Ml, e.Nl))
+                if pk == 0:
+                    A = gen_matrix(N)
+                    for lti in range(e.Ml // v):
+                        for ltj in range(e.Nl // v):
+                            gi, gj = lti * Px + pi, ltj * Py + pj
+                            loc[lti*v:(lti+1)*v, ltj*v:(ltj+1)*v] = \
+                                A[gi*v:(gi+1)*v, gj*v:(gj+1)*v]
+                e.set_matrix_local(loc)
+            else:
+                e.init_matrix(42)
             ms = e.factor()
         resid = e.validate()  # distributed collective (rank 0 -> broadcast)
         F = e.get_F_local()

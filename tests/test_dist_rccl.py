@@ -34,7 +34,8 @@ def _dist_env(tmp_path):
     return env
 
 
-def _run_ranks(tmp_path, N, v, Px, Py, Pz, reps=1, timeout=600):
+def _run_ranks(tmp_path, N, v, Px, Py, Pz, reps=1, timeout=600,
+               set_matrix=False):
     if not os.path.exists(SHIM):
         pytest.skip("shimccl.so not built (make -C tests)")
     P = Px * Py * Pz
@@ -46,7 +47,7 @@ def _run_ranks(tmp_path, N, v, Px, Py, Pz, reps=1, timeout=600):
         procs.append(subprocess.Popen(
             [sys.executable, os.path.join(HERE, "dist_worker.py"),
              str(N), str(v), str(Px), str(Py), str(Pz), str(r), str(reps),
-             out],
+             out] + (["set_matrix"] if set_matrix else []),
             env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
             text=True))
     logs = []
@@ -123,6 +124,21 @@ def test_dist_parity_vs_oracle(tmp_path, grid, N, v, reps, ftol):
         perm_sim = e.get_perm()
     assert np.array_equal(perm_sim, results[0]["perm"])
     assert np.array_equal(Fsim, F), "sim and distributed factors must be bit-identical"
+
+
+def test_dist_set_matrix_local(tmp_path):
+    """The documented drop-in data path (caller-supplied tile-cyclic local
+    slices through conflux_lu_set_matrix_local) in a real multi-process
+    world: same matrix as the generator, so the same oracle pins it."""
+    from oracle import Params, gen_matrix, lu_oracle
+
+    N, v, Px, Py, Pz = 1024, 128, 2, 2, 1
+    results = _run_ranks(tmp_path, N, v, Px, Py, Pz, set_matrix=True)
+    ref = lu_oracle(gen_matrix(N), Params(N, v, Px, Py, Pz))
+    for res in results:
+        assert np.array_equal(res["perm"], ref["perm"])
+    F = _assemble_F(results, N, v, Px, Py, Pz)
+    assert np.abs(F - ref["F"]).max() < 1e-11
 
 
 def test_bench_dist_launch(tmp_path):
