@@ -1248,10 +1248,11 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
     bool async_look = look && !c.sim && c.panel_stream &&
                       (lk ? atoi(lk) != 0 : true);
     // While the panel of step k+1 runs concurrently, cap the trailing
-    // update's grid so whole CUs stay free for it: a panel block needs
-    // 135 KB LDS (a full CU), so an uncapped GEMM flood starves the panel
-    // until the queue drains (measured 256 ms vs 251 ms sequential at
-    // N=16384 before this cap).  Default leaves 40 of 256 CUs free.
+    // update's grid so whole CUs stay free for it: an uncapped GEMM flood
+    // starves the panel until the queue drains (measured 256 ms vs 251 ms
+    // sequential at N=16384 before this cap).  Panel blocks need 67 KB LDS
+    // (256x256 shape, 2 blocks/CU); the default cap leaves 40 of 256 CUs
+    // free = up to 80 co-resident panel blocks.
     int gcap = 0;
     if (async_look) {
         static int env_cap = -1;
