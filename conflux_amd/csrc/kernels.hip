@@ -211,12 +211,14 @@ __global__ void k_row_move(const double *__restrict__ src, int64_t lds,
 //   region, publishes both affected rows sc1 + flag; blocks then finish from
 //   registers + the published slab (no plain re-reads of swapped data).
 #define PANEL_TPB 256
-#define PANEL_RPB 512           // rows per block (2 per thread)
+// (rows per block is now a template parameter of k_panel_factor —
+// QR*TPB, default 256x256 = 1 row/thread from the r02 shape sweep;
+// panel_shape() below picks the instantiation)
 #define PANEL_NB 32             // sub-panel width == register column budget
 
 // --------------------------------------------------------------------------
 // persistent sub-panel factorization (v2): ONE launch factors a whole
-// nb(<=32)-wide sub-panel.  Each block owns PANEL_RPB rows, held in LDS for
+// nb(<=32)-wide sub-panel.  Each block owns QR*TPB rows, held in LDS for
 // all nb columns; the only cross-block traffic per column is the slab
 // handshake: every block publishes its candidate row (full nb values) and
 // the block owning the diagonal row publishes it, so the row swap is
@@ -310,7 +312,7 @@ __global__ __launch_bounds__(256) void k_trsm_left_lower_unit32(
 
 // ---------------------------------------------------------------------------
 // persistent sub-panel getrf: ONE launch factors the nb(<=32)-wide sub-panel
-// with LAPACK partial pivoting (first-max rule).  Blocks own PANEL_RPB rows
+// with LAPACK partial pivoting (first-max rule).  Blocks own QR*TPB rows
 // held in LDS for all nb columns; per column the only cross-block traffic is
 // the slab handshake; the row swap is performed by the OWNING blocks from
 // slab-published rows (no cross-block matrix reads, no write races).
