@@ -1263,11 +1263,15 @@ int run_step(Ctx &c, int k, StepPlan &sp) {
                                           // > 448 > 464 at N=16384
         }
         // Only ranks that RUN step k+1's panel factor need the cap (and
-        // the residency guard); ranks outside column k+1 overlap only the
-        // panel chain's comm and keep the full-width GEMM.
+        // the residency guard): pj == ncol AND pk == 0 (the factor runs on
+        // layer 0 only — conflux_opt.hpp:689).  Other ranks overlap only
+        // the panel chain's comm and keep the full-width GEMM (r02 fix:
+        // pk > 0 ranks on 1x1xPz grids were capping for a factor they
+        // never run).
         int panel_rows = 0;
         for (auto &r : c.rs)
-            if (r.pj == ncol) panel_rows = std::max(panel_rows, r.nact);
+            if (r.pj == ncol && r.pk == 0)
+                panel_rows = std::max(panel_rows, r.nact);
         if (panel_rows > 0) {
             gcap = env_cap;
             // Residency guard: the persistent capped GEMM holds its CUs
