@@ -41,3 +41,17 @@ def test_create_rounds_n_up(lib):
         lib.conflux_lu_dims(h, *[ctypes.byref(x) for x in d])
         assert d[5].value == 256  # N_padded
         lib.conflux_lu_destroy(h)
+
+
+def test_header_symbols_all_exported(lib):
+    """Every symbol include/conflux_lu.h declares is exported by the built
+    library (no compute — pure dlsym checks)."""
+    import re
+    import os
+    hdr = open(os.path.join(os.path.dirname(__file__), "..", "include",
+                            "conflux_lu.h")).read()
+    decls = sorted(set(re.findall(r"\b(conflux_\w+)\s*\(", hdr)))
+    assert len(decls) >= 15
+    missing = [d for d in decls if not hasattr(lib, d)]
+    assert not missing, f"undefined in libconflux_lu.so: {missing}"
+    assert b"MI355X" in lib.conflux_lu_build_info()
