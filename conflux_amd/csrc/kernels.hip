@@ -1612,8 +1612,9 @@ int launch_panel_factor(double *panel, int64_t ldp, int m, int nb, void *sync,
     static int backoff = -1;
     if (backoff < 0) {
         const char *e = getenv("CONFLUX_PANEL_SLEEP");
-        backoff = e ? atoi(e) : 1;  // r02 matrix: s_sleep(1) in the key poll
-                                    // measured best-or-equal in context
+        backoff = e ? atoi(e) : 2;  // r02 sweep at the 256x256 shape:
+                                    // 0/1/2/4 -> 203.5/202.2/201.5/201.5
+                                    // ms/step in context
     }
     int qr, tpb;
     panel_shape(&qr, &tpb);
