@@ -21,9 +21,12 @@
 #include <sys/stat.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <map>
+#include <mutex>
 #include <string>
 #include <vector>
 
@@ -134,7 +137,11 @@ ncclResult_t stage_h2d(const void *host, void *dev, size_t bytes,
     return ncclSuccess;
 }
 
+ncclResult_t flush_ops_async();
+bool shim_async();
+
 ncclResult_t flush_ops() {
+    if (shim_async()) return flush_ops_async();
     // sends first (stage + publish), then recvs — NCCL group semantics:
     // no send blocks on its matching recv
     for (auto &op : g_ops) {
@@ -164,6 +171,122 @@ ncclResult_t flush_ops() {
 ncclResult_t enqueue(Op op) {
     g_ops.push_back(op);
     if (g_depth == 0) return flush_ops();
+    return ncclSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// SHIMCCL_ASYNC=1: stream-enqueued transport via hipLaunchHostFunc, matching
+// real RCCL's completion semantics — send/recv return immediately and the
+// blocking exchange happens when the STREAM reaches the host function, so
+// the engine's multi-stream choreography (dual-comm lookahead, event
+// gating) is exercised with true asynchronous ordering instead of the
+// host-blocking default.  Data stages through a pinned-buffer pool; a
+// callback that times out or hits I/O failure abort()s the rank (loud).
+// Collectives (allreduce/broadcast) stay host-blocking: they sit on barrier
+// paths where the host is about to synchronize anyway.
+// ---------------------------------------------------------------------------
+struct SlotPool {
+    std::mutex m;
+    std::multimap<size_t, void *> free_;
+    void *get(size_t n) {
+        {
+            std::lock_guard<std::mutex> g(m);
+            auto it = free_.lower_bound(n);
+            if (it != free_.end() && it->first <= 2 * n + 4096) {
+                void *p = it->second;
+                free_.erase(it);
+                return p;
+            }
+        }
+        void *p = nullptr;
+        if (hipHostMalloc(&p, n ? n : 8) != hipSuccess) return nullptr;
+        return p;
+    }
+    void put(void *p, size_t n) {
+        std::lock_guard<std::mutex> g(m);
+        free_.emplace(n, p);
+    }
+};
+SlotPool g_pool;
+
+struct HostOp {
+    std::string path;
+    size_t bytes;
+    void *slot;
+    bool remove;
+};
+
+void cb_publish(void *ud) {  // send: slot -> file, release slot
+    auto *h = (HostOp *)ud;
+    if (write_atomic(h->path, h->slot, h->bytes)) {
+        std::fprintf(stderr, "[shimccl async] publish failed %s\n",
+                     h->path.c_str());
+        abort();
+    }
+    g_pool.put(h->slot, h->bytes);
+    delete h;
+}
+
+void cb_wait_read(void *ud) {  // recv: file -> slot (slot released later)
+    auto *h = (HostOp *)ud;
+    if (read_blocking(h->path, h->slot, h->bytes, h->remove)) {
+        std::fprintf(stderr, "[shimccl async] recv failed %s\n",
+                     h->path.c_str());
+        abort();
+    }
+    delete h;
+}
+
+void cb_release(void *ud) {
+    auto *h = (HostOp *)ud;
+    g_pool.put(h->slot, h->bytes);
+    delete h;
+}
+
+bool shim_async() {
+    static int v = -1;
+    if (v < 0) {
+        const char *e = getenv("SHIMCCL_ASYNC");
+        v = e ? atoi(e) : 0;
+    }
+    return v != 0;
+}
+
+ncclResult_t flush_ops_async() {
+    // sends first, then recvs (NCCL group semantics), each as a chain of
+    // stream ops: no host blocking here at all
+    for (auto &op : g_ops) {
+        if (!op.is_send) continue;
+        void *slot = g_pool.get(op.bytes);
+        if (!slot) return ncclSystemError;
+        const uint64_t seq = op.comm->sseq[op.peer]++;
+        if (op.bytes &&
+            hipMemcpyAsync(slot, op.sbuf, op.bytes, hipMemcpyDeviceToHost,
+                           op.stream) != hipSuccess)
+            return ncclUnhandledCudaError;
+        auto *h = new HostOp{msgpath(op.comm, op.comm->rank, op.peer, seq),
+                             op.bytes, slot, false};
+        if (hipLaunchHostFunc(op.stream, cb_publish, h) != hipSuccess)
+            return ncclUnhandledCudaError;
+    }
+    for (auto &op : g_ops) {
+        if (op.is_send) continue;
+        void *slot = g_pool.get(op.bytes);
+        if (!slot) return ncclSystemError;
+        const uint64_t seq = op.comm->rseq[op.peer]++;
+        auto *h = new HostOp{msgpath(op.comm, op.peer, op.comm->rank, seq),
+                             op.bytes, slot, true};
+        if (hipLaunchHostFunc(op.stream, cb_wait_read, h) != hipSuccess)
+            return ncclUnhandledCudaError;
+        if (op.bytes &&
+            hipMemcpyAsync(op.rbuf, slot, op.bytes, hipMemcpyHostToDevice,
+                           op.stream) != hipSuccess)
+            return ncclUnhandledCudaError;
+        auto *h2 = new HostOp{"", op.bytes, slot, false};
+        if (hipLaunchHostFunc(op.stream, cb_release, h2) != hipSuccess)
+            return ncclUnhandledCudaError;
+    }
+    g_ops.clear();
     return ncclSuccess;
 }
 
