@@ -2398,6 +2398,42 @@ int conflux_lu_debug_dgemm_nt_bench(int M, int64_t N, int K, int iters,
     return CONFLUX_LU_OK;
 }
 
+/* achieved-GB/s probe for the row-shuffle kernels (report
+ * the HBM-bound kernels separately): gathers `rows` pivot rows of `cols`
+ * doubles out of an Ml x cols matrix through k_row_gather, `iters` times.
+ * Reported bytes = read + write of the moved rows. */
+int conflux_lu_debug_rowmove_bench(int Ml, int64_t cols, int rows, int iters,
+                                   double *gbps) {
+    hipStream_t s;
+    HIPCHK(hipStreamCreate(&s));
+    double *src, *dst;
+    int *idx;
+    HIPCHK(hipMalloc(&src, i64(Ml) * cols * 8));
+    HIPCHK(hipMalloc(&dst, i64(rows) * cols * 8));
+    HIPCHK(hipMalloc(&idx, rows * 4));
+    launch_init_matrix(src, Ml, (int)cols, (int)cols, 1, 1, 0, 0, 0, 11, s);
+    std::vector<int> h(rows);
+    for (int i = 0; i < rows; ++i)
+        h[i] = (int)((i64(i) * 2654435761u) % Ml);  // scattered rows
+    HIPCHK(hipMemcpy(idx, h.data(), rows * 4, hipMemcpyHostToDevice));
+    launch_row_gather(src, cols, dst, cols, idx, rows, cols, s);  // warmup
+    hipEvent_t a, b;
+    HIPCHK(hipEventCreate(&a));
+    HIPCHK(hipEventCreate(&b));
+    HIPCHK(hipEventRecord(a, s));
+    for (int i = 0; i < iters; ++i)
+        launch_row_gather(src, cols, dst, cols, idx, rows, cols, s);
+    HIPCHK(hipEventRecord(b, s));
+    HIPCHK(hipEventSynchronize(b));
+    float ms = 0;
+    HIPCHK(hipEventElapsedTime(&ms, a, b));
+    *gbps = 2.0 * rows * (double)cols * 8 * iters / (ms * 1e-3) / 1e9;
+    (void)hipFree(src); (void)hipFree(dst); (void)hipFree(idx);
+    (void)hipEventDestroy(a); (void)hipEventDestroy(b);
+    (void)hipStreamDestroy(s);
+    return CONFLUX_LU_OK;
+}
+
 int conflux_lu_debug_getrf(int n, int v, double *panel, int *ipiv_out) {
     Ctx c;
     c.v = v;
