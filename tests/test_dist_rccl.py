@@ -193,17 +193,20 @@ def test_bench_dist_launch(tmp_path):
     assert d["value"] > 0 and d["ms_per_step"] > 0
 
 
-@pytest.mark.parametrize("grid", ["2,2,1", "2,2,2"])
-def test_cholesky_dist(tmp_path, grid):
+@pytest.mark.parametrize("grid,async_mode", [
+    ("2,2,1", False), ("2,2,2", False), ("2,2,2", True),
+])
+def test_cholesky_dist(tmp_path, grid, async_mode):
     """The CONFCHOX distributed branches (depth reduce, L_kk column
     broadcast, slab + transpose spreads) executed multi-process through
-    shimccl, with the distributed Cholesky validation."""
+    shimccl — also under the async stream-enqueued transport — with the
+    distributed Cholesky validation."""
     chol = os.path.join(REPO, "conflux_amd", "cholesky_miniapp")
     if not os.path.exists(SHIM):
         pytest.skip("shimccl.so not built (make -C tests)")
     if not os.path.exists(chol):
         pytest.skip("cholesky_miniapp not built")
-    env = _dist_env(tmp_path)
+    env = _dist_env(tmp_path, async_mode)
     env["CONFLUX_SPAWN_OVERSUBSCRIBE"] = "1"
     out = subprocess.run(
         [chol, "--dim", "2048", "--tile", "256", "--grid", grid,
