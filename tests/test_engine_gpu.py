@@ -436,3 +436,28 @@ def test_lu_parity_8192_tournament(eng):
     assert np.array_equal(perm, r["perm"]), "pivots bit-exact at N=8192"
     assert np.abs(F - r["F"]).max() < 1e-9
     assert residual_check(A, perm, F) < 1e-13
+
+
+@pytest.mark.parametrize("N,v,Px,Py,Pz", [
+    (96, 32, 1, 1, 1),    # odd tile count (Nt=3)
+    (224, 32, 1, 1, 2),   # odd Nt=7 with depth replication
+    (192, 32, 2, 2, 1),   # Nt=6: odd tiles per rank row (3)
+    (192, 32, 2, 2, 2),   # same with Pz
+    (96, 8, 1, 1, 4),     # nlayr = v/Pz = 2 (deep replication, tiny slabs)
+    (160, 16, 2, 2, 2),   # Nt=10, small v
+    (384, 64, 2, 2, 4),   # v=64 split 4 ways (nlayr=16)
+])
+def test_lu_parity_envelope_corners(eng, N, v, Px, Py, Pz):
+    """Envelope corners the main GRIDS list misses: odd global tile counts,
+    odd tiles-per-rank, deep Pz splits with tiny nlayr slabs."""
+    A = gen_matrix(N)
+    r = lu_oracle(A, Params(N, v, Px, Py, Pz))
+    with eng.Engine(N, v, Px, Py, Pz, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor()
+        perm = e.get_perm()
+        F = e.get_F_global()
+    assert np.array_equal(perm, r["perm"])
+    assert np.abs(F - r["F"]).max() < TOL_F
+    assert residual_check(A, perm, F) < TOL_RES
