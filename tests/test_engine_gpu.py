@@ -264,6 +264,9 @@ CHOL_GRIDS = [
     (128, 16, 2, 2, 2),
     (128, 8, 4, 4, 2),
     (256, 32, 2, 2, 2),
+    (96, 32, 1, 1, 1),    # odd tile count (Nt=3)
+    (192, 32, 2, 2, 2),   # odd tiles per rank row
+    (96, 8, 1, 1, 4),     # nlayr = 2
 ]
 
 
