@@ -334,11 +334,20 @@ int factor_panel(Ctx &c, RankState &r, int n, std::vector<int> &ipiv_out) {
                 launch_trsm_left_lower_unit32(r.panel + i64(jb) * v + jb, v,
                                               r.panel + i64(jb) * v + jb + nb,
                                               v, nb, v - jb - nb, c.stream);
-                // trailing sub-panel update
+                // trailing sub-panel update.  CONFLUX_GLUE_CAP: under the
+                // capped trailing GEMM only ~45 workgroup slots are free;
+                // a full-grid glue launch churns through them in waves, a
+                // small persistent grid holds them (0 = uncapped)
+                static int glue_cap = -1;
+                if (glue_cap < 0) {
+                    const char *gc = getenv("CONFLUX_GLUE_CAP");
+                    glue_cap = gc ? atoi(gc) : 0;
+                }
                 launch_dgemm_f64(r.panel + i64(jb + nb) * v + jb, v,
                                  r.panel + i64(jb) * v + jb + nb, v,
                                  r.panel + i64(jb + nb) * v + jb + nb, v,
-                                 m - nb, v - jb - nb, nb, c.stream);
+                                 m - nb, v - jb - nb, nb, c.stream,
+                                 glue_cap);
             }
         }
     }
