@@ -334,10 +334,11 @@ int factor_panel(Ctx &c, RankState &r, int n, std::vector<int> &ipiv_out) {
                 launch_trsm_left_lower_unit32(r.panel + i64(jb) * v + jb, v,
                                               r.panel + i64(jb) * v + jb + nb,
                                               v, nb, v - jb - nb, c.stream);
-                // trailing sub-panel update.  CONFLUX_GLUE_CAP: under the
-                // capped trailing GEMM only ~45 workgroup slots are free;
-                // a full-grid glue launch churns through them in waves, a
-                // small persistent grid holds them (0 = uncapped)
+                // trailing sub-panel update.  CONFLUX_GLUE_CAP (measured
+                // ablation, default 0 = uncapped): a small persistent
+                // glue grid (48/64/96) measures 222/213/210 ms/step vs
+                // 199.5 uncapped — the full-grid launch's wave churn
+                // through the free slots beats holding them
                 static int glue_cap = -1;
                 if (glue_cap < 0) {
                     const char *gc = getenv("CONFLUX_GLUE_CAP");
