@@ -464,3 +464,22 @@ def test_lu_parity_envelope_corners(eng, N, v, Px, Py, Pz):
     assert np.array_equal(perm, r["perm"])
     assert np.abs(F - r["F"]).max() < TOL_F
     assert residual_check(A, perm, F) < TOL_RES
+
+
+def test_lu_parity_bench_size(eng):
+    """Pivots bit-exact vs the oracle restatement at the FULL bench size
+    (N=16384, v=512) — the oracle runs in ~25 s on the GPU box's host
+    cores, so the headline configuration itself is pivot-pinned every
+    round, not just residual-checked."""
+    N, v = 16384, 512
+    A = gen_matrix(N)
+    ref = lu_oracle(A, Params(N, v, 1, 1, 1))
+    with eng.Engine(N, v, 1, 1, 1, rank=-1) as e:
+        e.store_factors(True)
+        e.set_matrix_global(A)
+        e.factor()
+        perm = e.get_perm()
+        F = e.get_F_global()
+    assert np.array_equal(perm, ref["perm"]), "bench-size pivots"
+    assert np.abs(F - ref["F"]).max() < 1e-8
+    assert residual_check(A, perm, F) < 1e-13
