@@ -36,6 +36,7 @@ def _load():
     lib.conflux_lu_set_matrix_sim.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                               ctypes.c_void_p]
     lib.conflux_lu_store_factors.argtypes = [ctypes.c_void_p, ctypes.c_int]
+    lib.conflux_lu_set_pivoting.argtypes = [ctypes.c_void_p, ctypes.c_int]
     lib.conflux_lu_factor.argtypes = [ctypes.c_void_p,
                                       ctypes.POINTER(ctypes.c_double)]
     lib.conflux_lu_get_factors.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
@@ -144,6 +145,11 @@ class Engine:
 
     def store_factors(self, enable):
         _chk(lib().conflux_lu_store_factors(self._h, int(enable)), "store")
+
+    def set_pivoting(self, mode):
+        """1 = tournament (default), 0 = none (EmptyPivot fast path for
+        diagonally dominant inputs; perm stays identity)."""
+        _chk(lib().conflux_lu_set_pivoting(self._h, int(mode)), "pivoting")
 
     def factor(self):
         ms = ctypes.c_double()

@@ -111,6 +111,8 @@ struct Ctx {
     bool sim = false;            // all ranks in this process, 1 GPU
     bool input_dirty = true;     // A11 holds fresh input not yet snapshotted
     bool store_factors = true;
+    int pivoting = 1;            // 1 = tournament (reference), 0 = none
+                                 // (EmptyPivot fast path)
     bool have_comm = false;
     ncclComm_t comm{};
     ncclComm_t pcomm{};          // second comm for the lookahead panel chain
@@ -1874,6 +1876,337 @@ int chol_loop(Ctx &c, double *elapsed_ms) {
 }  // namespace
 
 // ===========================================================================
+// No-pivot LU (SURVEY §8f4 — the Python prototype's EmptyPivot strategy,
+// python/conflux.py; the C++ reference implements tournament only, so this
+// is parity-pinned against a numpy restatement instead): for diagonally
+// dominant inputs the whole tournament/row-movement machinery drops out and
+// row activation is static like Cholesky.  Per tile column k:
+//   n0  depth-reduce the k-th tile column (rows of tiles >= k) to layer 0
+//   n1  blocked in-place LU of the diagonal v x v tile (no pivoting);
+//       broadcast the packed LU (A00) to every rank
+//   n2  A10[below diag] <- A10 * U(A00)^-1 on the column ranks; C8 spread
+//   n3  A01[diag-tile rows, cols > k] depth-reduced, <- L(A00)^-1 * A01 on
+//       the row ranks; C9 spread
+//   n4  one rectangular trailing GEMM per rank (no masking: LU updates the
+//       full trailing block)
+// perm stays identity, Fres mirrors the pivoted layout, so
+// conflux_lu_validate works unchanged.
+// ===========================================================================
+namespace {
+
+int nopiv_step(Ctx &c, int k) {
+    c.acomm = c.comm;
+    const int v = c.v, Px = c.Px, Py = c.Py, Pz = c.Pz;
+    const int64_t Nl = c.Nl;
+    const int kcol = k % Py, krow = k % Px;
+    const int64_t loff = i64(k / Py) * v;
+    auto fnp_of = [&](int pi) { return v * ntiles_lt(c, pi, k); };
+    auto f2_of = [&](int pi) { return fnp_of(pi) + (pi == krow ? v : 0); };
+    auto ltj0_of = [&](int pj) { return (pj <= k) ? (k - pj) / Py + 1 : 0; };
+
+    // ---- n0: copy the k-th tile column into A10, depth-reduce ------------
+    for (auto &r : c.rs) {
+        if (r.pj != kcol) continue;
+        const int f = fnp_of(r.pi);
+        launch_copy2d(r.A11 + i64(f) * Nl + loff, Nl, r.A10 + i64(f) * v, v,
+                      c.Ml - f, v, c.stream);
+    }
+    if (Pz > 1) {
+        if (!c.sim) NCCLCHK(ncclGroupStart());
+        for (int pi = 0; pi < Px; ++pi) {
+            if (!c.sim && (c.rs[0].pi != pi || c.rs[0].pj != kcol)) continue;
+            const int f = fnp_of(pi);
+            auto buf = [&, f](RankState &x) { return x.A10 + i64(f) * v; };
+            if (reduce_over_pk(c, pi, kcol, i64(c.Ml - f) * v, buf))
+                return CONFLUX_LU_ECOMM;
+        }
+        if (!c.sim) {
+            NCCLCHK(ncclGroupEnd());
+            RankState &me = c.rs[0];
+            if (me.pj == kcol) {
+                const int f = fnp_of(me.pi);
+                auto buf = [&, f](RankState &x) { return x.A10 + i64(f) * v; };
+                if (reduce_over_pk_finish(c, me.pi, kcol, i64(c.Ml - f) * v,
+                                          buf))
+                    return CONFLUX_LU_ECOMM;
+            }
+        }
+    }
+
+    // ---- n1: blocked no-pivot LU of the diagonal tile; broadcast A00 -----
+    {
+        RankState *own = get_rs(c, krow, kcol, 0);
+        if (own) {
+            double *T = own->A10 + i64(fnp_of(krow)) * v;
+            size_t slot;
+            if (ev_begin(c, 1, 0, &slot)) return CONFLUX_LU_EHIP;
+            const int NB = conflux_panel_nb();
+            for (int jb = 0; jb < v; jb += NB) {
+                const int nb = std::min(NB, v - jb);
+                launch_getrf32_nopiv(T + i64(jb) * v + jb, v, nb, c.stream);
+                if (jb + nb < v) {
+                    launch_trsm_right_upper32(T + i64(jb) * v + jb, v,
+                                              T + i64(jb + nb) * v + jb, v,
+                                              nb, v - jb - nb, 0, c.stream);
+                    launch_trsm_left_lower_unit32(T + i64(jb) * v + jb, v,
+                                                  T + i64(jb) * v + jb + nb,
+                                                  v, nb, v - jb - nb,
+                                                  c.stream);
+                    launch_dgemm_f64(T + i64(jb + nb) * v + jb, v,
+                                     T + i64(jb) * v + jb + nb, v,
+                                     T + i64(jb + nb) * v + jb + nb, v,
+                                     v - jb - nb, v - jb - nb, nb, c.stream);
+                }
+            }
+            if (ev_end(c, slot)) return CONFLUX_LU_EHIP;
+            launch_copy2d(T, v, own->A00, v, v, v, c.stream);
+        }
+        // broadcast the packed LU to every rank (both halves are consumed)
+        if (c.sim) {
+            RankState &src = *get_rs(c, krow, kcol, 0);
+            for (auto &d : c.rs)
+                if (&d != &src)
+                    if (d2d(c, d.A00, src.A00, i64(v) * v))
+                        return CONFLUX_LU_EHIP;
+        } else if (c.world > 1) {
+            RankState &me = c.rs[0];
+            NCCLCHK(ncclGroupStart());
+            if (own && own->grank == me.grank) {
+                for (int g = 0; g < c.world; ++g)
+                    if (g != me.grank)
+                        NCCLCHK(ncclSend(me.A00, i64(v) * v, ncclDouble, g,
+                                         c.comm, c.stream));
+            } else {
+                NCCLCHK(ncclRecv(me.A00, i64(v) * v, ncclDouble,
+                                 grank_of(c, krow, kcol, 0), c.comm,
+                                 c.stream));
+            }
+            NCCLCHK(ncclGroupEnd());
+        }
+    }
+
+    // ---- n2: A10 <- A10 U^-1 below the diagonal; store L; C8 spread ------
+    for (auto &r : c.rs) {
+        if (r.pj != kcol || r.pk != 0) continue;
+        const int f2 = f2_of(r.pi);
+        const int n2 = c.Ml - f2;
+        if (n2 > 0 && trsm_right_upper(c, r, r.A10 + i64(f2) * v, v, n2))
+            return CONFLUX_LU_EINTERNAL;
+        if (c.store_factors) {
+            if (n2 > 0)
+                launch_copy2d(r.A10 + i64(f2) * v, v,
+                              r.Fres + i64(f2) * Nl + loff, Nl, n2, v,
+                              c.stream);
+            if (r.pi == krow)  // diagonal tile: packed LU
+                launch_copy2d(r.A00, v, r.Fres + i64(fnp_of(krow)) * Nl + loff,
+                              Nl, v, v, c.stream);
+        }
+    }
+    if (Py == 1 && Pz == 1) {
+        for (auto &r : c.rs) {
+            const int f2 = f2_of(r.pi);
+            launch_copy2d(r.A10 + i64(f2) * v, v, r.A10Rcv, c.nlayr,
+                          c.Ml - f2, v, c.stream);
+        }
+    } else {
+        for (int pi = 0; pi < Px; ++pi) {
+            RankState *root = get_rs(c, pi, kcol, 0);
+            const int f2 = f2_of(pi);
+            const int n2 = c.Ml - f2;
+            if (c.sim) {
+                if (n2 > 0)
+                    launch_slab_pack(root->A10 + i64(f2) * v, v, n2, c.nlayr,
+                                     Pz, root->slabs, c.stream);
+                for (int pj = 0; pj < Py; ++pj)
+                    for (int pk = 0; pk < Pz; ++pk) {
+                        RankState &d = *get_rs(c, pi, pj, pk);
+                        if (n2 > 0 &&
+                            d2d(c, d.A10Rcv,
+                                root->slabs + i64(pk) * n2 * c.nlayr,
+                                i64(n2) * c.nlayr))
+                            return CONFLUX_LU_EHIP;
+                    }
+            } else {
+                RankState &me = c.rs[0];
+                if (me.pi != pi || n2 <= 0) continue;
+                NCCLCHK(ncclGroupStart());
+                if (root && root->grank == me.grank) {
+                    launch_slab_pack(me.A10 + i64(f2) * v, v, n2, c.nlayr, Pz,
+                                     me.slabs, c.stream);
+                    for (int pj = 0; pj < Py; ++pj)
+                        for (int pk = 0; pk < Pz; ++pk) {
+                            if (pj == kcol && pk == 0) continue;
+                            NCCLCHK(ncclSend(me.slabs + i64(pk) * n2 * c.nlayr,
+                                             i64(n2) * c.nlayr, ncclDouble,
+                                             grank_of(c, pi, pj, pk), c.comm,
+                                             c.stream));
+                        }
+                } else {
+                    NCCLCHK(ncclRecv(me.A10Rcv, i64(n2) * c.nlayr, ncclDouble,
+                                     grank_of(c, pi, kcol, 0), c.comm,
+                                     c.stream));
+                }
+                NCCLCHK(ncclGroupEnd());
+                if (root && root->grank == me.grank)
+                    if (d2d(c, me.A10Rcv, me.slabs, i64(n2) * c.nlayr))
+                        return CONFLUX_LU_EHIP;
+            }
+        }
+    }
+
+    // ---- n3: A01 row panel (diag-tile rows, cols > k): reduce, solve,
+    // store U, C9 spread ----------------------------------------------------
+    for (auto &r : c.rs) {
+        if (r.pi != krow) continue;
+        const int64_t c0 = i64(ltj0_of(r.pj)) * v;
+        const int64_t w = Nl - c0;
+        if (w <= 0) continue;
+        launch_copy2d(r.A11 + i64(fnp_of(krow)) * Nl + c0, Nl, r.A01, w, v, w,
+                      c.stream);
+    }
+    if (Pz > 1) {
+        if (!c.sim) NCCLCHK(ncclGroupStart());
+        for (int pj = 0; pj < Py; ++pj) {
+            if (!c.sim && (c.rs[0].pi != krow || c.rs[0].pj != pj)) continue;
+            const int64_t w = Nl - i64(ltj0_of(pj)) * v;
+            if (w <= 0) continue;
+            auto buf = [](RankState &x) { return x.A01; };
+            if (reduce_over_pk(c, krow, pj, i64(c.v) * w, buf))
+                return CONFLUX_LU_ECOMM;
+        }
+        if (!c.sim) {
+            NCCLCHK(ncclGroupEnd());
+            RankState &me = c.rs[0];
+            if (me.pi == krow) {
+                const int64_t w = Nl - i64(ltj0_of(me.pj)) * v;
+                auto buf = [](RankState &x) { return x.A01; };
+                if (w > 0 &&
+                    reduce_over_pk_finish(c, krow, me.pj, i64(c.v) * w, buf))
+                    return CONFLUX_LU_ECOMM;
+            }
+        }
+    }
+    for (auto &r : c.rs) {
+        if (r.pi != krow || r.pk != 0) continue;
+        const int64_t c0 = i64(ltj0_of(r.pj)) * v;
+        const int64_t w = Nl - c0;
+        if (w <= 0) continue;
+        if (trsm_left_lower(c, r, r.A01, w, w)) return CONFLUX_LU_EINTERNAL;
+        if (c.store_factors)
+            launch_copy2d(r.A01, w, r.Fres + i64(fnp_of(krow)) * Nl + c0, Nl,
+                          v, w, c.stream);
+    }
+    // C9: spread the solved row panel down each column (into A01Rcv at the
+    // SAME local column offset, ld Nl)
+    for (int pj = 0; pj < Py; ++pj) {
+        const int64_t c0 = i64(ltj0_of(pj)) * v;
+        const int64_t w = Nl - c0;
+        if (w <= 0) continue;
+        RankState *root = get_rs(c, krow, pj, 0);
+        if (c.sim) {
+            for (int pi = 0; pi < Px; ++pi)
+                for (int pk = 0; pk < Pz; ++pk) {
+                    RankState &d = *get_rs(c, pi, pj, pk);
+                    launch_copy2d(root->A01 + i64(pk) * c.nlayr * w, w,
+                                  d.A01Rcv + c0, Nl, c.nlayr, w, c.stream);
+                }
+        } else {
+            RankState &me = c.rs[0];
+            if (me.pj != pj) continue;
+            NCCLCHK(ncclGroupStart());
+            if (root && root->grank == me.grank) {
+                for (int pi = 0; pi < Px; ++pi)
+                    for (int pk = 0; pk < Pz; ++pk) {
+                        if (pi == krow && pk == 0) continue;
+                        NCCLCHK(ncclSend(me.A01 + i64(pk) * c.nlayr * w,
+                                         i64(c.nlayr) * w, ncclDouble,
+                                         grank_of(c, pi, pj, pk), c.comm,
+                                         c.stream));
+                    }
+            } else {
+                NCCLCHK(ncclRecv(me.redtmp, i64(c.nlayr) * w, ncclDouble,
+                                 grank_of(c, krow, pj, 0), c.comm, c.stream));
+            }
+            NCCLCHK(ncclGroupEnd());
+            if (root && root->grank == me.grank)
+                launch_copy2d(me.A01, w, me.A01Rcv + c0, Nl, c.nlayr, w,
+                              c.stream);
+            else
+                launch_copy2d(me.redtmp, w, me.A01Rcv + c0, Nl, c.nlayr, w,
+                              c.stream);
+        }
+    }
+
+    // ---- n4: one rectangular trailing update per rank ---------------------
+    for (auto &r : c.rs) {
+        const int rstart = f2_of(r.pi);
+        const int64_t c0 = i64(ltj0_of(r.pj)) * v;
+        const int M2 = c.Ml - rstart;
+        const int64_t N2 = Nl - c0;
+        if (M2 <= 0 || N2 <= 0) continue;
+        const double fl = 2.0 * M2 * (double)N2 * c.nlayr;
+        size_t slot;
+        if (ev_begin(c, 0, fl, &slot)) return CONFLUX_LU_EHIP;
+        launch_dgemm_f64(r.A10Rcv, c.nlayr, r.A01Rcv + c0, Nl,
+                         r.A11 + i64(rstart) * Nl + c0, Nl, M2, N2, c.nlayr,
+                         c.stream);
+        if (ev_end(c, slot)) return CONFLUX_LU_EHIP;
+    }
+    return 0;
+}
+
+int nopiv_loop(Ctx &c, double *elapsed_ms) {
+    for (auto &r : c.rs) {
+        if (c.store_factors) {
+            if (ensure_factor_bufs(c, r, /*need_hist=*/false))
+                return CONFLUX_LU_EHIP;
+            launch_zero2d(r.Fres, c.Nl, c.Ml, c.Nl, c.stream);
+        }
+    }
+    c.pivotInds.resize(c.M);
+    std::iota(c.pivotInds.begin(), c.pivotInds.end(), 0);
+    c.evs_used = 0;
+    for (auto &t : c.cats) t = TimeCat{};
+    HIPCHK(hipStreamSynchronize(c.stream));
+    if (c.have_comm) {
+        static double *dummy = nullptr;
+        if (!dummy) HIPCHK(hipMalloc(&dummy, 8));
+        NCCLCHK(ncclAllReduce(dummy, dummy, 1, ncclDouble, ncclSum, c.comm,
+                              c.stream));
+        HIPCHK(hipStreamSynchronize(c.stream));
+    }
+    const auto t1 = std::chrono::high_resolution_clock::now();
+    for (int k = 0; k < c.Nt; ++k) {
+        int rc = nopiv_step(c, k);
+        if (rc) return rc;
+    }
+    HIPCHK(hipStreamSynchronize(c.stream));
+    if (c.have_comm) {
+        static double *dummy2 = nullptr;
+        if (!dummy2) HIPCHK(hipMalloc(&dummy2, 8));
+        NCCLCHK(ncclAllReduce(dummy2, dummy2, 1, ncclDouble, ncclSum, c.comm,
+                              c.stream));
+        HIPCHK(hipStreamSynchronize(c.stream));
+    }
+    const auto t2 = std::chrono::high_resolution_clock::now();
+    HIPCHK(hipGetLastError());
+    if (elapsed_ms)
+        *elapsed_ms =
+            std::chrono::duration<double, std::milli>(t2 - t1).count();
+    for (size_t i = 0; i < c.evs_used; ++i) {
+        float ms = 0;
+        HIPCHK(hipEventElapsedTime(&ms, c.evs[i].a, c.evs[i].b));
+        TimeCat &t = c.cats[c.evs[i].cat];
+        t.seconds += ms * 1e-3;
+        t.launches += 1;
+        t.flops += c.evs[i].flops;
+    }
+    return 0;
+}
+
+}  // namespace
+
+// ===========================================================================
 // C ABI
 // ===========================================================================
 struct conflux_lu_ctx : Ctx {};
@@ -2065,6 +2398,17 @@ int conflux_lu_store_factors(conflux_lu_ctx *c, int enable) {
     return CONFLUX_LU_OK;
 }
 
+/* Pivoting strategy: 1 = tournament pivoting (the reference
+ * algorithm, default); 0 = NO pivoting — the Python prototype's EmptyPivot
+ * fast path for diagonally dominant inputs (python/conflux.py; the C++
+ * reference implements tournament only).  With mode 0 the permutation is
+ * identity and the tournament/row-movement machinery drops out. */
+int conflux_lu_set_pivoting(conflux_lu_ctx *c, int mode) {
+    if (mode != 0 && mode != 1) return CONFLUX_LU_EARG;
+    c->pivoting = mode;
+    return CONFLUX_LU_OK;
+}
+
 // The reference's LU_rep factors a COPY — lu_params::data survives the call
 // (conflux_opt.hpp:398).  Same semantics here: the first factor() after a
 // matrix upload snapshots A11; later factor() calls restore the snapshot.
@@ -2088,7 +2432,8 @@ static int snapshot_or_restore(conflux_lu_ctx *c) {
 
 int conflux_lu_factor(conflux_lu_ctx *c, double *elapsed_ms) {
     if (snapshot_or_restore(c)) return CONFLUX_LU_EHIP;
-    int rc = factor_loop(*c, elapsed_ms);
+    int rc = c->pivoting ? factor_loop(*c, elapsed_ms)
+                         : nopiv_loop(*c, elapsed_ms);
     if (rc) {
         std::fprintf(stderr, "[conflux_lu] factor failed: %s\n",
                      c->err.c_str());

@@ -1449,6 +1449,33 @@ __global__ __launch_bounds__(256) void k_potrf32(double *__restrict__ A,
         A[(i / nb) * lda + i % nb] = sA[i / nb][i % nb];
 }
 
+// 32x32 LU WITHOUT pivoting (the Python prototype's EmptyPivot fast path,
+// SURVEY §8f4 — for diagonally dominant inputs): per column, scale below
+// the diagonal by 1/pivot and rank-1 the trailing sub-block.  Single block,
+// same shape as k_potrf32.
+__global__ __launch_bounds__(256) void k_getrf32_nopiv(double *__restrict__ A,
+                                                       int64_t lda, int nb) {
+    __shared__ double sA[32][33];
+    const int tid = threadIdx.x;
+    for (int i = tid; i < nb * nb; i += 256)
+        sA[i / nb][i % nb] = A[(i / nb) * lda + i % nb];
+    __syncthreads();
+    for (int c = 0; c < nb; ++c) {
+        const double piv = sA[c][c];
+        const double inv = (piv != 0.0) ? 1.0 / piv : 0.0;  // LAPACK-style:
+        __syncthreads();                                    // no scaling on 0
+        if (tid > c && tid < nb && piv != 0.0) sA[tid][c] *= inv;
+        __syncthreads();
+        for (int e = tid; e < nb * nb; e += 256) {
+            const int i = e / nb, j = e % nb;
+            if (i > c && j > c) sA[i][j] -= sA[i][c] * sA[c][j];
+        }
+        __syncthreads();
+    }
+    for (int i = tid; i < nb * nb; i += 256)
+        A[(i / nb) * lda + i % nb] = sA[i / nb][i % nb];
+}
+
 // ---------------------------------------------------------------------------
 // misc small kernels for the distributed path
 // ---------------------------------------------------------------------------
@@ -2072,6 +2099,11 @@ void launch_frob2(const double *A, int64_t nelem, double *out, hipStream_t s) {
 void launch_potrf32(double *A, int64_t lda, int nb, hipStream_t s) {
     if (nb <= 0) return;
     hipLaunchKernelGGL(k_potrf32, dim3(1), dim3(256), 0, s, A, lda, nb);
+}
+
+void launch_getrf32_nopiv(double *A, int64_t lda, int nb, hipStream_t s) {
+    if (nb <= 0) return;
+    hipLaunchKernelGGL(k_getrf32_nopiv, dim3(1), dim3(256), 0, s, A, lda, nb);
 }
 
 int g_dgemm_variant = -1;  // 0 = 4-wave, 1 = 8-wave; env CONFLUX_GEMM_VARIANT

@@ -56,6 +56,7 @@ void launch_tril(const double *F, double *L, int64_t n, hipStream_t s);
 void launch_triu(const double *F, double *U, int64_t n, hipStream_t s);
 void launch_frob2(const double *A, int64_t nelem, double *out, hipStream_t s);
 void launch_potrf32(double *A, int64_t lda, int nb, hipStream_t s);
+void launch_getrf32_nopiv(double *A, int64_t lda, int nb, hipStream_t s);
 void launch_dgemm_f64_nt(const double *A, int64_t lda, const double *B,
                          int64_t ldb, double *C, int64_t ldc, int M, int64_t N,
                          int K, hipStream_t s);

@@ -18,5 +18,5 @@
 #   - the reference's own LU loop compiled in-container (oracle/_ref,
 #     recipe under oracle/ref_build/) on small grids under mpiexec.
 # ============================================================================
-from .lu_oracle import Params, lu_oracle, residual_check
+from .lu_oracle import Params, lu_oracle, lu_nopivot, residual_check
 from .gen_input import gen_matrix

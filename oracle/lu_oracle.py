@@ -350,3 +350,33 @@ def residual_check(A: np.ndarray, perm: np.ndarray, F: np.ndarray) -> float:
     Uu = np.triu(F)
     PA = A[perm[:N]]
     return float(np.linalg.norm(PA - L @ Uu) / np.linalg.norm(A))
+
+
+def lu_nopivot(A: np.ndarray, v: int) -> np.ndarray:
+    """Blocked LU WITHOUT pivoting — the Python prototype's EmptyPivot
+    strategy (python/conflux.py pivoting enum; the C++ reference implements
+    tournament only), restated as the parity target for the engine's
+    no-pivot fast path (conflux_lu_set_pivoting(ctx, 0)).  Requires a
+    diagonally dominant input.  Returns F: strict lower = L (unit diag
+    implied), upper = U; the permutation is identity by construction."""
+    import scipy.linalg as la
+    N = A.shape[0]
+    assert N % v == 0
+    F = np.array(A, dtype=np.float64)
+    for k0 in range(0, N, v):
+        k1 = k0 + v
+        T = F[k0:k1, k0:k1]
+        for c in range(v):
+            piv = T[c, c]
+            if piv != 0.0:
+                T[c + 1:, c] /= piv
+            T[c + 1:, c + 1:] -= np.outer(T[c + 1:, c], T[c, c + 1:])
+        if k1 < N:
+            L = np.tril(T, -1) + np.eye(v)
+            U = np.triu(T)
+            F[k0:k1, k1:] = la.solve_triangular(
+                L, F[k0:k1, k1:], lower=True, unit_diagonal=True)
+            F[k1:, k0:k1] = la.solve_triangular(
+                U, F[k1:, k0:k1].T, trans='T').T
+            F[k1:, k1:] -= F[k1:, k0:k1] @ F[k0:k1, k1:]
+    return F

@@ -483,3 +483,33 @@ def test_lu_parity_bench_size(eng):
     assert np.array_equal(perm, ref["perm"]), "bench-size pivots"
     assert np.abs(F - ref["F"]).max() < 1e-8
     assert residual_check(A, perm, F) < 1e-13
+
+
+# ---------------- no-pivot fast path (EmptyPivot) -------------
+
+@pytest.mark.parametrize("N,v,Px,Py,Pz", [
+    (256, 64, 1, 1, 1),
+    (256, 64, 2, 2, 1),
+    (256, 64, 2, 2, 2),
+    (512, 128, 2, 2, 1),
+])
+def test_lu_nopivot_parity(eng, N, v, Px, Py, Pz):
+    """conflux_lu_set_pivoting(ctx, 0): the EmptyPivot fast path on a
+    diagonally dominant input, vs the numpy no-pivot restatement; identity
+    permutation and the unchanged device residual."""
+    from oracle import lu_nopivot
+    A = gen_matrix(N)
+    S = 0.5 * (A + A.T) + 2.0 * N * np.eye(N)  # == init_matrix_spd fill
+    ref = lu_nopivot(S, v)
+    with eng.Engine(N, v, Px, Py, Pz, rank=-1) as e:
+        e.store_factors(True)
+        e.set_pivoting(0)
+        e.set_matrix_global(S)
+        e.factor()
+        perm = e.get_perm()
+        F = e.get_F_global()
+        resid = e.validate()
+    assert np.array_equal(perm, np.arange(N))
+    scale = np.abs(ref).max()
+    assert np.abs(F - ref).max() < 1e-11 * scale
+    assert resid < 1e-14
