@@ -8,7 +8,7 @@ standing in for RCCL (all ranks share one GPU).  With SHIMCCL_DIR fixed,
 ncclGetUniqueId is deterministic, so every rank derives the same uid locally
 and no out-of-band exchange is needed.
 
-usage: dist_worker.py N v Px Py Pz rank reps out.npz [set_matrix]
+usage: dist_worker.py N v Px Py Pz rank reps out.npz [set_matrix|nopiv]
 
 With the optional 9th arg "set_matrix", the rank uploads its tile-cyclic
 local slice through conflux_lu_set_matrix_local (the documented drop-in
@@ -30,11 +30,18 @@ def main():
 
     P = Px * Py * Pz
     uid = Engine.make_uid()
-    use_set = len(sys.argv) > 9 and sys.argv[9] == "set_matrix"
+    mode = sys.argv[9] if len(sys.argv) > 9 else ""
+    use_set = mode == "set_matrix"
     with Engine(N, v, Px, Py, Pz, rank=rank, world=P, uid=uid) as e:
         e.store_factors(True)
+        if mode == "nopiv":
+            e.set_pivoting(0)
         ms = 0.0
         for _ in range(reps):
+            if mode == "nopiv":
+                e.init_matrix_spd(42)  # diagonally dominant input
+                ms = e.factor()
+                continue
             if use_set:
                 from oracle import gen_matrix
                 pi, pj, pk = rank // (Py * Pz), (rank // Pz) % Py, rank % Pz

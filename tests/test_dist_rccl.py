@@ -39,7 +39,7 @@ def _dist_env(tmp_path, async_mode=False):
 
 
 def _run_ranks(tmp_path, N, v, Px, Py, Pz, reps=1, timeout=600,
-               set_matrix=False, async_mode=False):
+               set_matrix=False, async_mode=False, mode=None):
     if not os.path.exists(SHIM):
         pytest.skip("shimccl.so not built (make -C tests)")
     P = Px * Py * Pz
@@ -51,7 +51,8 @@ def _run_ranks(tmp_path, N, v, Px, Py, Pz, reps=1, timeout=600,
         procs.append(subprocess.Popen(
             [sys.executable, os.path.join(HERE, "dist_worker.py"),
              str(N), str(v), str(Px), str(Py), str(Pz), str(r), str(reps),
-             out] + (["set_matrix"] if set_matrix else []),
+             out] + (["set_matrix"] if set_matrix else [mode] if mode
+                     else []),
             env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
             text=True))
     logs = []
@@ -166,6 +167,24 @@ def test_dist_set_matrix_local(tmp_path):
         assert np.array_equal(res["perm"], ref["perm"])
     F = _assemble_F(results, N, v, Px, Py, Pz)
     assert np.abs(F - ref["F"]).max() < 1e-11
+
+
+def test_dist_nopivot(tmp_path):
+    """The no-pivot fast path through the real multi-process
+    distributed branches: identity permutation, factors vs the numpy
+    no-pivot restatement on the SPD (diagonally dominant) generator fill."""
+    from oracle import gen_matrix, lu_nopivot
+
+    N, v, Px, Py, Pz = 1024, 128, 2, 2, 2
+    results = _run_ranks(tmp_path, N, v, Px, Py, Pz, mode="nopiv")
+    A = gen_matrix(N)
+    S = 0.5 * (A + A.T) + 2.0 * N * np.eye(N)
+    ref = lu_nopivot(S, v)
+    for res in results:
+        assert np.array_equal(res["perm"], np.arange(N))
+        assert float(res["resid"]) < 1e-14
+    F = _assemble_F(results, N, v, Px, Py, Pz)
+    assert np.abs(F - ref).max() < 1e-11 * np.abs(ref).max()
 
 
 def test_bench_dist_launch(tmp_path):
