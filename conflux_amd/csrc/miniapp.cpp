@@ -27,12 +27,16 @@ static void usage() {
         "  -l, --print_limit l   (accepted for compatibility)\n"
         "  -t, --type t          weak|strong|other (label only)\n"
         "      --sim             all ranks in one process on one GPU\n"
-        "      --timing          skip factor collection (bench mode)\n");
+        "      --timing          skip factor collection (bench mode)\n"
+        "      --pivoting M      tournament (default) | none (EmptyPivot\n"
+        "                        fast path; uses the SPD diagonally\n"
+        "                        dominant generator fill)\n");
 }
 
 int main(int argc, char **argv) {
     int N = 1000, b = 256, reps = 2, Px = 0, Py = 0, Pz = 0;
     std::string type = "other";
+    std::string pivoting = "tournament";
     bool sim = false, timing = false;
     for (int i = 1; i < argc; ++i) {
         std::string a = argv[i];
@@ -48,6 +52,7 @@ int main(int argc, char **argv) {
         else if (a == "-t" || a.rfind("--type", 0) == 0) type = val("t");
         else if (a == "--sim") sim = true;
         else if (a == "--timing") timing = true;
+        else if (a.rfind("--pivoting", 0) == 0) pivoting = val("M");
         else if (a == "-p" || a.rfind("--p_grid", 0) == 0) {
             std::string g = val("p");
             if (std::sscanf(g.c_str(), "%d,%d,%d", &Px, &Py, &Pz) != 3) {
@@ -95,6 +100,8 @@ int main(int argc, char **argv) {
         return 1;
     }
     conflux_lu_store_factors(ctx, timing ? 0 : 1);
+    const bool nopiv = pivoting == "none";
+    if (nopiv) conflux_lu_set_pivoting(ctx, 0);
 
     const bool print0 = (rank <= 0);
     if (print0) {
@@ -112,7 +119,9 @@ int main(int argc, char **argv) {
     const int sqrtP = (int)std::max(1.0, std::floor(std::sqrt((double)P)));
     const int N_base = (type == "weak") ? N / sqrtP : N;
     for (int i = 0; i < reps + 1; ++i) {
-        conflux_lu_init_matrix(ctx, 42);
+        // no-pivot requires a diagonally dominant input: use the SPD fill
+        if (nopiv) conflux_lu_init_matrix_spd(ctx, 42);
+        else conflux_lu_init_matrix(ctx, 42);
         double ms = 0;
         rc = conflux_lu_factor(ctx, &ms);
         if (rc) { std::fprintf(stderr, "factor failed rc=%d\n", rc); return 1; }

@@ -51,3 +51,19 @@ def test_cholesky_single_rank_validated():
                   out.stdout)
     assert m, out.stdout
     assert float(m.group(1)) < 1e-13
+
+
+def test_miniapp_nopivot_validated():
+    """CLI surface of the no-pivot fast path: --pivoting none factors the
+    diagonally dominant SPD fill and the device residual validates it."""
+    if not os.path.exists(MINIAPP):
+        pytest.skip("conflux_miniapp not built")
+    out = subprocess.run(
+        [MINIAPP, "-N", "2048", "-b", "256", "--p_grid=1,1,1", "-r", "1",
+         "--pivoting", "none"],
+        capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stdout + out.stderr
+    m = re.search(r"relative residual \|\|PA-LU\|\|_F/\|\|A\|\|_F = (\S+)",
+                  out.stdout)
+    assert m, out.stdout
+    assert float(m.group(1)) < 1e-14
